@@ -1,0 +1,187 @@
+"""Native Llama-family implementation on the framework's CDNA4 ops.
+
+This is the flagship model used by bench.py (BASELINE.json: Llama-2-7B FSDP
+bf16). It is written directly against torchacc_amd.ops — RMSNorm, fused RoPE,
+flash attention, SwiGLU, fused-linear-cross-entropy — with the projection
+GEMMs on hipBLASLt via torch.matmul. HF-transformers models are supported
+separately through the patch layer (utils/patch.py); this native module
+avoids HF overhead on the hot path.
+
+Context parallelism: when a CP mode is set (Config.dist.sp), attention
+dispatches to ulysses / ring / 2D FlashSequence and the inputs are expected
+sequence-sharded (bench/accelerate handle the sharding).
+"""
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from ..ops.cross_entropy import linear_cross_entropy
+from ..ops.flash_attn import flash_attn_xla
+from ..ops.rmsnorm import RMSNorm
+from ..ops.rope import apply_rotary_pos_emb, build_rope_cache
+from ..ops.swiglu import swiglu
+
+
+@dataclass
+class LlamaConfig:
+    vocab_size: int = 32000
+    hidden_size: int = 4096
+    intermediate_size: int = 11008
+    num_hidden_layers: int = 32
+    num_attention_heads: int = 32
+    num_key_value_heads: int = 32
+    max_position_embeddings: int = 4096
+    rms_norm_eps: float = 1e-5
+    rope_theta: float = 10000.0
+    tie_word_embeddings: bool = False
+    # context-parallel attention mode: None|'ulysses'|'ring'|'2d'
+    cp_mode: Optional[str] = None
+
+
+def llama_2_7b(**kw) -> "LlamaConfig":
+    return LlamaConfig(**kw)
+
+
+def llama_2_70b(**kw) -> "LlamaConfig":
+    return LlamaConfig(
+        hidden_size=8192, intermediate_size=28672, num_hidden_layers=80,
+        num_attention_heads=64, num_key_value_heads=8, **kw)
+
+
+def llama_tiny(**kw) -> "LlamaConfig":
+    """4-layer toy config (driver config #1: tiny DP=1 CPU plumbing)."""
+    return LlamaConfig(
+        vocab_size=1024, hidden_size=256, intermediate_size=688,
+        num_hidden_layers=4, num_attention_heads=8, num_key_value_heads=8,
+        max_position_embeddings=512, **kw)
+
+
+class LlamaAttention(nn.Module):
+
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        h, hk = cfg.num_attention_heads, cfg.num_key_value_heads
+        self.head_dim = cfg.hidden_size // h
+        self.q_proj = nn.Linear(cfg.hidden_size, h * self.head_dim,
+                                bias=False)
+        self.k_proj = nn.Linear(cfg.hidden_size, hk * self.head_dim,
+                                bias=False)
+        self.v_proj = nn.Linear(cfg.hidden_size, hk * self.head_dim,
+                                bias=False)
+        self.o_proj = nn.Linear(h * self.head_dim, cfg.hidden_size,
+                                bias=False)
+
+    def forward(self, x, cos, sin):
+        b, s, _ = x.shape
+        h, hk = self.cfg.num_attention_heads, self.cfg.num_key_value_heads
+        q = self.q_proj(x).view(b, s, h, self.head_dim)
+        k = self.k_proj(x).view(b, s, hk, self.head_dim)
+        v = self.v_proj(x).view(b, s, hk, self.head_dim)
+        mode = self.cfg.cp_mode
+        if mode is None:
+            q, k = apply_rotary_pos_emb(q, k, cos, sin)
+            o = flash_attn_xla(q, k, v, causal=True)
+        elif mode == "ulysses":
+            from ..ops.context_parallel import ulysses
+
+            def rope_fn(qq, kk):
+                return apply_rotary_pos_emb(qq, kk, cos, sin)
+
+            o = ulysses(q, k, v, causal=True, rope_func=rope_fn)
+        elif mode == "ring":
+            from ..ops.context_parallel import (get_inter_cp_group,
+                                                ring_attention)
+            import torch.distributed as dist
+            g = get_inter_cp_group()
+            r = dist.get_rank(g) if g is not None else 0
+            q, k = apply_rotary_pos_emb(q, k, cos[r * s:(r + 1) * s],
+                                        sin[r * s:(r + 1) * s])
+            o = ring_attention(q, k, v, causal=True)
+        elif mode == "2d":
+            from ..ops.context_parallel import (context_parallel_2d,
+                                                get_context_parallel_group)
+            import torch.distributed as dist
+            g = get_context_parallel_group()
+            r = dist.get_rank(g) if g is not None else 0
+            q, k = apply_rotary_pos_emb(q, k, cos[r * s:(r + 1) * s],
+                                        sin[r * s:(r + 1) * s])
+            o = context_parallel_2d(q, k, v, causal=True)
+        else:
+            raise ValueError(f"unknown cp mode {mode}")
+        return self.o_proj(o.reshape(b, s, h * self.head_dim))
+
+
+class LlamaMLP(nn.Module):
+
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.gate_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size,
+                                   bias=False)
+        self.up_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size,
+                                 bias=False)
+        self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size,
+                                   bias=False)
+
+    def forward(self, x):
+        return self.down_proj(swiglu(self.gate_proj(x), self.up_proj(x)))
+
+
+class LlamaDecoderLayer(nn.Module):
+
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+        self.self_attn = LlamaAttention(cfg)
+        self.post_attention_layernorm = RMSNorm(cfg.hidden_size,
+                                                cfg.rms_norm_eps)
+        self.mlp = LlamaMLP(cfg)
+
+    def forward(self, x, cos, sin):
+        x = x + self.self_attn(self.input_layernorm(x), cos, sin)
+        x = x + self.mlp(self.post_attention_layernorm(x))
+        return x
+
+
+class LlamaForCausalLM(nn.Module):
+
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.config = cfg
+        self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+        self.layers = nn.ModuleList(
+            [LlamaDecoderLayer(cfg) for _ in range(cfg.num_hidden_layers)])
+        self.norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False)
+        if cfg.tie_word_embeddings:
+            self.lm_head.weight = self.embed_tokens.weight
+        head_dim = cfg.hidden_size // cfg.num_attention_heads
+        cos, sin = build_rope_cache(cfg.max_position_embeddings, head_dim,
+                                    cfg.rope_theta)
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+        self.apply(self._init_weights)
+
+    @staticmethod
+    def _init_weights(m):
+        if isinstance(m, nn.Linear):
+            nn.init.normal_(m.weight, std=0.02)
+        elif isinstance(m, nn.Embedding):
+            nn.init.normal_(m.weight, std=0.02)
+
+    def forward(self, input_ids: torch.Tensor,
+                labels: Optional[torch.Tensor] = None,
+                attention_mask=None):
+        x = self.embed_tokens(input_ids)
+        cos, sin = self.rope_cos, self.rope_sin
+        for layer in self.layers:
+            x = layer(x, cos, sin)
+        x = self.norm(x)
+        if labels is not None:
+            # shift: predict token t+1 from position t; fused linear+CE
+            hs = x[:, :-1, :].reshape(-1, x.shape[-1])
+            tg = labels[:, 1:].reshape(-1)
+            return linear_cross_entropy(hs, self.lm_head.weight, tg)
+        return self.lm_head(x)

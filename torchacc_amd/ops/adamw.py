@@ -1,0 +1,98 @@
+"""Fused AdamW for MI355X (+ syncfree found_inf gating).
+
+Replaces torch_xla.amp.syncfree.AdamW (reference utils/patch.py:55-57): the
+HIP kernel performs the whole AdamW update for a list of params in one
+multi-tensor launch, keeps fp32 exp_avg/exp_avg_sq (and optional fp32 master
+weights for bf16/fp16 params), and accepts a device-side ``found_inf`` flag:
+when nonzero the update is a no-op, so fp16 loss scaling never host-syncs.
+"""
+import math
+from typing import Optional
+
+import torch
+
+from ._backend import dispatch
+
+
+class AdamW(torch.optim.Optimizer):
+
+    def __init__(self, params, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
+                 weight_decay=1e-2, use_master_weights: bool = True):
+        defaults = dict(lr=lr, betas=betas, eps=eps,
+                        weight_decay=weight_decay)
+        super().__init__(params, defaults)
+        self.use_master_weights = use_master_weights
+
+    @torch.no_grad()
+    def step(self, closure=None, found_inf: Optional[torch.Tensor] = None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        for group in self.param_groups:
+            beta1, beta2 = group["betas"]
+            lr = group["lr"]
+            eps = group["eps"]
+            wd = group["weight_decay"]
+            params, grads, exp_avgs, exp_avg_sqs, masters, steps = \
+                [], [], [], [], [], []
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = torch.zeros(
+                        (), dtype=torch.float32, device=p.device)
+                    state["exp_avg"] = torch.zeros_like(
+                        p, dtype=torch.float32)
+                    state["exp_avg_sq"] = torch.zeros_like(
+                        p, dtype=torch.float32)
+                    if self.use_master_weights and p.dtype in (
+                            torch.bfloat16, torch.float16):
+                        state["master"] = p.detach().float().clone()
+                params.append(p)
+                grads.append(p.grad)
+                exp_avgs.append(state["exp_avg"])
+                exp_avg_sqs.append(state["exp_avg_sq"])
+                masters.append(state.get("master"))
+                steps.append(state["step"])
+            if not params:
+                continue
+            ext = dispatch(params[0])
+            if ext is not None:
+                fi = found_inf if found_inf is not None else \
+                    torch.zeros((), dtype=torch.float32,
+                                device=params[0].device)
+                master_list = [
+                    m if m is not None else torch.empty(0) for m in masters
+                ]
+                ext.fused_adamw(params, grads, exp_avgs, exp_avg_sqs,
+                                master_list, steps, fi, lr, beta1, beta2,
+                                eps, wd)
+            else:
+                if found_inf is not None and bool(found_inf != 0):
+                    continue
+                for p, g, m, v, mw, st in zip(params, grads, exp_avgs,
+                                              exp_avg_sqs, masters, steps):
+                    st += 1
+                    t = float(st)
+                    gf = g.float()
+                    m.mul_(beta1).add_(gf, alpha=1 - beta1)
+                    v.mul_(beta2).addcmul_(gf, gf, value=1 - beta2)
+                    bc1 = 1 - beta1 ** t
+                    bc2 = 1 - beta2 ** t
+                    denom = (v / bc2).sqrt_().add_(eps)
+                    upd = (m / bc1) / denom
+                    tgt = mw if mw is not None else p
+                    tgtf = tgt if tgt.dtype == torch.float32 else tgt.float()
+                    tgtf = tgtf.mul_(1 - lr * wd).add_(upd, alpha=-lr)
+                    if mw is not None:
+                        mw.copy_(tgtf)
+                        p.copy_(tgtf.to(p.dtype))
+                    else:
+                        p.copy_(tgtf.to(p.dtype))
+        return loss
+
+
+# torch_xla.amp.syncfree-compatible aliases
+SyncFreeAdamW = AdamW

@@ -1,0 +1,363 @@
+"""Flash-attention v2 ops on hand-written CDNA4 MFMA kernels.
+
+Reimplements the API surface of the reference's ops/flash_attn.py:11-601
+(FlashAttnXla / FlashAttnVarlenXla / FlashAttnVarlenPositionIdsXla and the
+user wrappers flash_attn_xla / flash_attn_varlen_xla /
+flash_attn_varlen_position_ids_xla) on a single eager ROCm backend:
+
+- fixed-length [b, s, h, d] fwd/bwd with causal, GQA (h_k | h) and sliding
+  window;
+- varlen via an int32 attention mask [b, s_k] (right-padding; per-batch
+  lengths are extracted and passed to the kernel);
+- packed-sequence varlen via position_ids [1, total] (cu_seqlens derived
+  from position-id resets, reference ops/flash_attn.py:173-218) — used by
+  the HF packed-sample training path;
+- explicit cu_seqlens varlen (used by ring attention).
+
+The softmax log-sum-exp is returned in fp32 [b, h, s] exactly like FA2 so the
+context-parallel LSE merge math is unchanged. fp16/bf16 only. Dropout is
+accepted but only p=0.0 is supported by the CDNA4 kernels so far.
+"""
+import math
+from typing import Optional, Tuple
+
+import torch
+
+from ._backend import dispatch
+
+
+def _check_qkv(q, k, v):
+    if q.is_cuda:
+        assert q.dtype in (torch.float16, torch.bfloat16), \
+            "flash attention requires fp16/bf16"
+    assert q.dtype == k.dtype == v.dtype
+    assert q.dim() == 4 and k.dim() == 4 and v.dim() == 4, \
+        "expected [batch, seqlen, heads, head_dim]"
+    assert k.shape == v.shape
+    assert q.shape[0] == k.shape[0] and q.shape[3] == k.shape[3]
+    assert q.shape[2] % k.shape[2] == 0, "GQA requires h_k | h_q"
+
+
+def _ref_attention(q, k, v, softmax_scale, causal, window, q_lens=None,
+                   k_lens=None):
+    """fp32 composite reference; returns (out, lse[b,h,sq])."""
+    b, sq, h, d = q.shape
+    sk = k.shape[1]
+    hk = k.shape[2]
+    rep = h // hk
+    qf = q.float().permute(0, 2, 1, 3)             # [b,h,sq,d]
+    kf = k.float().permute(0, 2, 1, 3)             # [b,hk,sk,d]
+    vf = v.float().permute(0, 2, 1, 3)
+    if rep > 1:
+        kf = kf.repeat_interleave(rep, dim=1)
+        vf = vf.repeat_interleave(rep, dim=1)
+    scores = torch.matmul(qf, kf.transpose(-1, -2)) * softmax_scale
+    neg = torch.finfo(torch.float32).min
+    iq = torch.arange(sq, device=q.device).view(1, 1, sq, 1)
+    ik = torch.arange(sk, device=q.device).view(1, 1, 1, sk)
+    # causal alignment: bottom-right (FA2 semantics)
+    if causal:
+        shift = sk - sq
+        scores = scores.masked_fill(ik > iq + shift, neg)
+    wl, wr = window
+    if wl >= 0:
+        shift = sk - sq
+        scores = scores.masked_fill(ik < iq + shift - wl, neg)
+    if wr >= 0 and not causal:
+        shift = sk - sq
+        scores = scores.masked_fill(ik > iq + shift + wr, neg)
+    if k_lens is not None:
+        klen = k_lens.view(b, 1, 1, 1)
+        scores = scores.masked_fill(ik >= klen, neg)
+    lse = torch.logsumexp(scores, dim=-1)          # [b,h,sq]
+    p = torch.exp(scores - lse.unsqueeze(-1))
+    p = torch.nan_to_num(p)                        # fully-masked rows
+    out = torch.matmul(p, vf)                      # [b,h,sq,d]
+    if q_lens is not None:
+        qmask = (iq.view(1, 1, sq, 1) < q_lens.view(b, 1, 1, 1))
+        out = out * qmask
+        lse = torch.where(qmask.squeeze(-1), lse,
+                          torch.zeros_like(lse))
+    return out.permute(0, 2, 1, 3).to(q.dtype), lse
+
+
+class FlashAttnFunc(torch.autograd.Function):
+
+    @staticmethod
+    def forward(ctx, q, k, v, dropout_p, softmax_scale, causal, window_size,
+                alibi_slopes, deterministic, q_lens, k_lens):
+        _check_qkv(q, k, v)
+        assert dropout_p == 0.0, \
+            "dropout is not yet supported by the CDNA4 kernels"
+        assert alibi_slopes is None, \
+            "alibi is not yet supported by the CDNA4 kernels"
+        if softmax_scale is None:
+            softmax_scale = 1.0 / math.sqrt(q.shape[-1])
+        q, k, v = [t.contiguous() for t in (q, k, v)]
+        ext = dispatch(q)
+        wl, wr = window_size
+        if ext is not None:
+            out, lse = ext.fa_forward(
+                q, k, v, softmax_scale, causal, wl, wr,
+                q_lens if q_lens is not None else torch.empty(0),
+                k_lens if k_lens is not None else torch.empty(0))
+        else:
+            out, lse = _ref_attention(q, k, v, softmax_scale, causal,
+                                      (wl, wr), q_lens, k_lens)
+        ctx.save_for_backward(
+            q, k, v, out, lse,
+            q_lens if q_lens is not None else torch.empty(0),
+            k_lens if k_lens is not None else torch.empty(0))
+        ctx.softmax_scale = softmax_scale
+        ctx.causal = causal
+        ctx.window = (wl, wr)
+        ctx.deterministic = deterministic
+        return out, lse
+
+    @staticmethod
+    def backward(ctx, dout, _dlse):
+        q, k, v, out, lse, q_lens, k_lens = ctx.saved_tensors
+        q_lens = q_lens if q_lens.numel() else None
+        k_lens = k_lens if k_lens.numel() else None
+        ext = dispatch(q)
+        dout = dout.contiguous()
+        wl, wr = ctx.window
+        if ext is not None:
+            dq, dk, dv = ext.fa_backward(
+                dout, q, k, v, out, lse, ctx.softmax_scale, ctx.causal,
+                wl, wr,
+                q_lens if q_lens is not None else torch.empty(0),
+                k_lens if k_lens is not None else torch.empty(0))
+        else:
+            dq, dk, dv = _ref_fa_backward(dout, q, k, v, out, lse,
+                                          ctx.softmax_scale, ctx.causal,
+                                          (wl, wr), q_lens, k_lens)
+        return (dq, dk, dv) + (None,) * 8
+
+
+def _ref_fa_backward(dout, q, k, v, out, lse, softmax_scale, causal, window,
+                     q_lens, k_lens):
+    """Recompute-based fp32 reference backward.
+
+    ``out``/``lse`` are the GLOBAL attention output / logsumexp: delta =
+    rowsum(dout * out) so the same routine serves both plain FA backward and
+    per-block ring-attention backward (where p is normalized by the global
+    lse and delta must come from the merged output)."""
+    b, sq, h, d = q.shape
+    sk, hk = k.shape[1], k.shape[2]
+    rep = h // hk
+    qf = q.float().permute(0, 2, 1, 3)
+    kf = k.float().permute(0, 2, 1, 3)
+    vf = v.float().permute(0, 2, 1, 3)
+    dof = dout.float().permute(0, 2, 1, 3)
+    if rep > 1:
+        kf = kf.repeat_interleave(rep, dim=1)
+        vf = vf.repeat_interleave(rep, dim=1)
+    scores = torch.matmul(qf, kf.transpose(-1, -2)) * softmax_scale
+    neg = torch.finfo(torch.float32).min
+    iq = torch.arange(sq, device=q.device).view(1, 1, sq, 1)
+    ik = torch.arange(sk, device=q.device).view(1, 1, 1, sk)
+    shift = sk - sq
+    if causal:
+        scores = scores.masked_fill(ik > iq + shift, neg)
+    wl, wr = window
+    if wl >= 0:
+        scores = scores.masked_fill(ik < iq + shift - wl, neg)
+    if wr >= 0 and not causal:
+        scores = scores.masked_fill(ik > iq + shift + wr, neg)
+    if k_lens is not None:
+        scores = scores.masked_fill(ik >= k_lens.view(b, 1, 1, 1), neg)
+    p = torch.exp(scores - lse.unsqueeze(-1))
+    p = torch.nan_to_num(p)
+    if q_lens is not None:
+        qmask = iq < q_lens.view(b, 1, 1, 1)
+        p = p * qmask
+        dof = dof * qmask
+    dv = torch.matmul(p.transpose(-1, -2), dof)
+    dp = torch.matmul(dof, vf.transpose(-1, -2))
+    delta = (dof * out.float().permute(0, 2, 1, 3)).sum(-1, keepdim=True)
+    ds = p * (dp - delta) * softmax_scale
+    dq = torch.matmul(ds, kf)
+    dk = torch.matmul(ds.transpose(-1, -2), qf)
+    if rep > 1:
+        dk = dk.view(b, hk, rep, sk, d).sum(2)
+        dv = dv.view(b, hk, rep, sk, d).sum(2)
+    return (dq.permute(0, 2, 1, 3).to(q.dtype),
+            dk.permute(0, 2, 1, 3).to(k.dtype),
+            dv.permute(0, 2, 1, 3).to(v.dtype))
+
+
+# ---------------------------------------------------------------------------
+# user wrappers (reference-compatible names)
+# ---------------------------------------------------------------------------
+
+def flash_attn_xla(q, k, v, dropout_p=0.0, softmax_scale=None, causal=False,
+                   window_size=(-1, -1), alibi_slopes=None,
+                   deterministic=False, return_attn_probs=False):
+    """Fixed-length flash attention, q/k/v [b, s, h, d].
+
+    Name kept from the reference for drop-in compatibility; runs on the
+    CDNA4 HIP kernel (there is no XLA here).
+    """
+    out, lse = FlashAttnFunc.apply(q, k, v, dropout_p, softmax_scale, causal,
+                                   window_size, alibi_slopes, deterministic,
+                                   None, None)
+    if return_attn_probs:
+        return out, lse, None
+    return out
+
+
+flash_attn_func = flash_attn_xla
+
+
+def flash_attn_varlen_xla(q, k, v, attention_mask=None, dropout_p=0.0,
+                          softmax_scale=None, causal=False,
+                          window_size=(-1, -1), alibi_slopes=None,
+                          deterministic=False, return_attn_probs=False):
+    """Varlen-by-mask flash attention (reference FlashAttnVarlenXla,
+    ops/flash_attn.py:219-262): ``attention_mask`` is an int mask [b, s_k]
+    marking valid keys (right padding)."""
+    k_lens = None
+    q_lens = None
+    if attention_mask is not None:
+        assert attention_mask.dim() == 2
+        k_lens = attention_mask.to(torch.int32).sum(-1).to(torch.int32)
+        if q.shape[1] == k.shape[1]:
+            q_lens = k_lens
+    out, lse = FlashAttnFunc.apply(q, k, v, dropout_p, softmax_scale, causal,
+                                   window_size, alibi_slopes, deterministic,
+                                   q_lens, k_lens)
+    if return_attn_probs:
+        return out, lse, None
+    return out
+
+
+def position_ids_to_cu_seqlens(position_ids: torch.Tensor) -> torch.Tensor:
+    """[1, total] packed position ids -> int32 cu_seqlens (resets at 0),
+    reference FlashAttnVarlenPositionIdsXla semantics."""
+    pos = position_ids.reshape(-1)
+    starts = torch.nonzero(pos == 0).reshape(-1)
+    total = pos.numel()
+    cu = torch.cat([starts.to(torch.int32),
+                    torch.tensor([total], dtype=torch.int32,
+                                 device=pos.device)])
+    return cu
+
+
+def flash_attn_varlen_position_ids_xla(q, k, v, position_ids, dropout_p=0.0,
+                                       softmax_scale=None, causal=False,
+                                       window_size=(-1, -1),
+                                       alibi_slopes=None,
+                                       deterministic=False,
+                                       return_attn_probs=False):
+    """Packed-sequence flash attention: bsz must be 1, sequences delimited by
+    position-id resets (reference ops/flash_attn.py:173-218, 488)."""
+    assert q.shape[0] == 1, "position-ids varlen requires batch size 1"
+    cu = position_ids_to_cu_seqlens(position_ids)
+    max_len = int((cu[1:] - cu[:-1]).max())
+    out, lse = FlashAttnVarlenFunc.apply(
+        q.squeeze(0), k.squeeze(0), v.squeeze(0), cu, cu, max_len, max_len,
+        dropout_p, softmax_scale, causal, window_size, deterministic)
+    out = out.unsqueeze(0)
+    if return_attn_probs:
+        return out, lse, None
+    return out
+
+
+class FlashAttnVarlenFunc(torch.autograd.Function):
+    """True packed varlen with explicit cu_seqlens (used by ring attention
+    and the position-ids path). q/k/v are [total, h, d]."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, cu_q, cu_k, max_q, max_k, dropout_p,
+                softmax_scale, causal, window_size, deterministic):
+        assert dropout_p == 0.0
+        if softmax_scale is None:
+            softmax_scale = 1.0 / math.sqrt(q.shape[-1])
+        q, k, v = [t.contiguous() for t in (q, k, v)]
+        ext = dispatch(q)
+        wl, wr = window_size
+        if ext is not None:
+            out, lse = ext.fa_varlen_forward(q, k, v, cu_q.to(q.device),
+                                             cu_k.to(q.device), max_q, max_k,
+                                             softmax_scale, causal, wl, wr)
+        else:
+            out, lse = _ref_varlen(q, k, v, cu_q, cu_k, softmax_scale,
+                                   causal, (wl, wr))
+        ctx.save_for_backward(q, k, v, out, lse, cu_q, cu_k)
+        ctx.meta = (max_q, max_k, softmax_scale, causal, window_size)
+        return out, lse
+
+    @staticmethod
+    def backward(ctx, dout, _dlse):
+        q, k, v, out, lse, cu_q, cu_k = ctx.saved_tensors
+        max_q, max_k, softmax_scale, causal, window_size = ctx.meta
+        ext = dispatch(q)
+        dout = dout.contiguous()
+        wl, wr = window_size
+        if ext is not None:
+            dq, dk, dv = ext.fa_varlen_backward(
+                dout, q, k, v, out, lse, cu_q, cu_k, max_q, max_k,
+                softmax_scale, causal, wl, wr)
+        else:
+            dq, dk, dv = _ref_varlen_backward(dout, q, k, v, out, lse, cu_q,
+                                              cu_k, softmax_scale, causal,
+                                              (wl, wr))
+        return (dq, dk, dv) + (None,) * 9
+
+
+def _ref_varlen(q, k, v, cu_q, cu_k, softmax_scale, causal, window):
+    total, h, d = q.shape
+    out = torch.zeros_like(q)
+    lse = torch.full((h, total), 0.0, dtype=torch.float32, device=q.device)
+    for i in range(cu_q.numel() - 1):
+        qs, qe = int(cu_q[i]), int(cu_q[i + 1])
+        ks, ke = int(cu_k[i]), int(cu_k[i + 1])
+        if qe == qs:
+            continue
+        o, l = _ref_attention(q[qs:qe].unsqueeze(0), k[ks:ke].unsqueeze(0),
+                              v[ks:ke].unsqueeze(0), softmax_scale, causal,
+                              window)
+        out[qs:qe] = o.squeeze(0)
+        lse[:, qs:qe] = l.squeeze(0)
+    return out, lse
+
+
+def _ref_varlen_backward(dout, q, k, v, out, lse, cu_q, cu_k, softmax_scale,
+                         causal, window):
+    dq = torch.zeros_like(q)
+    dk = torch.zeros_like(k)
+    dv = torch.zeros_like(v)
+    for i in range(cu_q.numel() - 1):
+        qs, qe = int(cu_q[i]), int(cu_q[i + 1])
+        ks, ke = int(cu_k[i]), int(cu_k[i + 1])
+        if qe == qs:
+            continue
+        dqi, dki, dvi = _ref_fa_backward(
+            dout[qs:qe].unsqueeze(0), q[qs:qe].unsqueeze(0),
+            k[ks:ke].unsqueeze(0), v[ks:ke].unsqueeze(0),
+            out[qs:qe].unsqueeze(0), lse[:, qs:qe].unsqueeze(0),
+            softmax_scale, causal, window, None, None)
+        dq[qs:qe] = dqi.squeeze(0)
+        dk[ks:ke] = dki.squeeze(0)
+        dv[ks:ke] = dvi.squeeze(0)
+    return dq, dk, dv
+
+
+def flash_attn_varlen_func(q, k, v, cu_seqlens_q, cu_seqlens_k, max_seqlen_q,
+                           max_seqlen_k, dropout_p=0.0, softmax_scale=None,
+                           causal=False, window_size=(-1, -1),
+                           alibi_slopes=None, deterministic=False,
+                           return_attn_probs=False):
+    out, lse = FlashAttnVarlenFunc.apply(q, k, v, cu_seqlens_q, cu_seqlens_k,
+                                         max_seqlen_q, max_seqlen_k,
+                                         dropout_p, softmax_scale, causal,
+                                         window_size, deterministic)
+    if return_attn_probs:
+        return out, lse, None
+    return out
+
+
+# SPMD-named alias kept for API compatibility (reference ops/flash_attn.py:66)
+def spmd_flash_attn_varlen_xla(*args, **kwargs):
+    return flash_attn_varlen_xla(*args, **kwargs)

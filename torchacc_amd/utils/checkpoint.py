@@ -1,0 +1,57 @@
+"""Gradient (activation) checkpointing (reference utils/checkpoint.py:17-81).
+
+Wraps modules whose class name is in ``gc_cls`` (or the root model when the
+set is empty) with torch.utils.checkpoint (non-reentrant: plays well with the
+FSDP engine's pre-backward re-gather hooks). ``gc_cnt`` limits how many
+instances are wrapped (reference fsdp.py:182-194 interleave semantics).
+"""
+from typing import Optional, Set
+
+import torch
+from torch.utils.checkpoint import checkpoint
+
+from .logger import logger
+
+
+class CheckpointWrapper(torch.nn.Module):
+
+    def __init__(self, mod: torch.nn.Module):
+        super().__init__()
+        self._checkpoint_wrapped_module = mod
+
+    def forward(self, *args, **kwargs):
+        if torch.is_grad_enabled():
+            return checkpoint(self._checkpoint_wrapped_module, *args,
+                              use_reentrant=False, **kwargs)
+        return self._checkpoint_wrapped_module(*args, **kwargs)
+
+    def __getattr__(self, name):
+        try:
+            return super().__getattr__(name)
+        except AttributeError:
+            return getattr(self._checkpoint_wrapped_module, name)
+
+
+def gradient_checkpoint(model: torch.nn.Module,
+                        gc_cls: Optional[Set[str]] = None,
+                        gc_cnt: Optional[int] = None) -> torch.nn.Module:
+    gc_cls = set(gc_cls or ())
+    if not gc_cls:
+        return CheckpointWrapper(model)
+    targets = []
+    for mod in model.modules():
+        if isinstance(mod, CheckpointWrapper):
+            continue
+        for name, child in mod.named_children():
+            if child.__class__.__name__ in gc_cls and \
+                    not isinstance(child, CheckpointWrapper):
+                targets.append((mod, name, child))
+    count = 0
+    for mod, name, child in targets:
+        if gc_cnt is not None and count >= gc_cnt:
+            break
+        setattr(mod, name, CheckpointWrapper(child))
+        count += 1
+    logger.info("gradient checkpointing: wrapped %d modules (%s)", count,
+                sorted(gc_cls))
+    return model

