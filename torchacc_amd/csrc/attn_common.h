@@ -1,0 +1,34 @@
+// Shared helpers for the CDNA4 attention kernels.
+#pragma once
+#include "common.h"
+
+#define LOG2E_F 1.4426950408889634f
+
+// D-matrix row for accumulator register r on lane-half hi (32x32 mfma):
+// row = (r&3) + 8*(r>>2) + 4*hi
+#define CROW(r, hi) (((r) & 3) + 8 * ((r) >> 2) + 4 * (hi))
+
+DEVINLINE unsigned attn_cvt_pk_bf16(float lo, float hi) {
+  unsigned r;
+  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
+               : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+// T12 redistribution: given a 32-wide f32 accumulator pair layout
+// D[crow(r,hi)][col=lane&31], produce the bf16x8 B/A-fragment for mfma step
+// tp (16 rows [16*tp, 16*tp+16)): lane half hi receives rows 16tp+8hi+j.
+// Returns 4 packed u32 (8 bf16) in frag[0..3].
+template <typename V16>
+DEVINLINE void t12_pack_frag(const V16& p16 /*16 regs*/, int tp,
+                             unsigned* frag) {
+#pragma unroll
+  for (int u = 0; u < 2; ++u) {
+    const int r = 2 * u + 8 * tp;
+    unsigned va = attn_cvt_pk_bf16(p16[r], p16[r + 1]);
+    unsigned vb = attn_cvt_pk_bf16(p16[r + 4], p16[r + 5]);
+    auto sw = __builtin_amdgcn_permlane32_swap(va, vb, false, false);
+    frag[u] = sw[0];
+    frag[u + 2] = sw[1];
+  }
+}

@@ -1,0 +1,337 @@
+#include "hip/hip_runtime.h"
+// Memory-bound fused elementwise/normalization kernels for gfx950:
+// RMSNorm fwd/bwd, RoPE (fused q+k), SwiGLU fwd/bwd.
+//
+// All HBM-bound: bf16x8 (16 B/lane) vectorized loads/stores, grid-stride
+// loops capped at ~2048 workgroups (guide G11/G13). fp32 math throughout.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// RMSNorm
+// ---------------------------------------------------------------------------
+// one 256-thread block per row (4 waves); H assumed multiple of 8.
+
+template <typename T>
+struct VecIO;
+
+// bf16 path: 8 elements per 16B
+__global__ void rmsnorm_fwd_kernel(const short* __restrict__ x,
+                                   const short* __restrict__ w,
+                                   short* __restrict__ y,
+                                   float* __restrict__ inv_rms, int rows,
+                                   int H, float eps) {
+  __shared__ float red[16];
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const short* xr = x + (long)row * H;
+    short* yr = y + (long)row * H;
+    float ss = 0.f;
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+      s16x8 v = *reinterpret_cast<const s16x8*>(xr + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf16_to_f32(v[j]);
+        ss += f * f;
+      }
+    }
+    ss = block_reduce_sum<4>(ss, red);
+    float r = rsqrtf(ss / H + eps);
+    if (threadIdx.x == 0) inv_rms[row] = r;
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+      s16x8 v = *reinterpret_cast<const s16x8*>(xr + i);
+      s16x8 wv = *reinterpret_cast<const s16x8*>(w + i);
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = f32_to_bf16(bf16_to_f32(v[j]) * r * bf16_to_f32(wv[j]));
+      *reinterpret_cast<s16x8*>(yr + i) = o;
+    }
+  }
+}
+
+// backward: dx per row + dw accumulated in LDS across this block's rows,
+// then one fp32 atomicAdd per element (guide G12: partial-reduce first).
+__global__ void rmsnorm_bwd_kernel(const short* __restrict__ dy,
+                                   const short* __restrict__ x,
+                                   const short* __restrict__ w,
+                                   const float* __restrict__ inv_rms,
+                                   short* __restrict__ dx,
+                                   float* __restrict__ dw,  // fp32 [H]
+                                   int rows, int H) {
+  extern __shared__ float smem[];  // [H] dw accumulator + 16 reduce
+  float* dw_acc = smem;
+  float* red = smem + H;
+  for (int i = threadIdx.x; i < H; i += blockDim.x) dw_acc[i] = 0.f;
+  __syncthreads();
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const short* xr = x + (long)row * H;
+    const short* dyr = dy + (long)row * H;
+    short* dxr = dx + (long)row * H;
+    const float r = inv_rms[row];
+    // pass 1: ddot = mean(w*dy*xhat)
+    float dot = 0.f;
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+      s16x8 xv = *reinterpret_cast<const s16x8*>(xr + i);
+      s16x8 dv = *reinterpret_cast<const s16x8*>(dyr + i);
+      s16x8 wv = *reinterpret_cast<const s16x8*>(w + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xh = bf16_to_f32(xv[j]) * r;
+        dot += bf16_to_f32(wv[j]) * bf16_to_f32(dv[j]) * xh;
+      }
+    }
+    dot = block_reduce_sum<4>(dot, red) / H;
+    // pass 2: dx = r*(w*dy - xhat*dot); dw_acc += dy*xhat
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+      s16x8 xv = *reinterpret_cast<const s16x8*>(xr + i);
+      s16x8 dv = *reinterpret_cast<const s16x8*>(dyr + i);
+      s16x8 wv = *reinterpret_cast<const s16x8*>(w + i);
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xh = bf16_to_f32(xv[j]) * r;
+        float dyf = bf16_to_f32(dv[j]);
+        o[j] = f32_to_bf16(r * (bf16_to_f32(wv[j]) * dyf - xh * dot) +
+                           0.f);
+        dw_acc[i + j] += dyf * xh;  // each thread owns cols i..i+7
+      }
+      *reinterpret_cast<s16x8*>(dxr + i) = o;
+    }
+    __syncthreads();
+  }
+  for (int i = threadIdx.x; i < H; i += blockDim.x)
+    atomicAdd(&dw[i], dw_acc[i]);
+}
+
+// fp32 variants (CPU-parity/debug path; also used when model is fp32)
+__global__ void rmsnorm_fwd_kernel_f32(const float* __restrict__ x,
+                                       const float* __restrict__ w,
+                                       float* __restrict__ y,
+                                       float* __restrict__ inv_rms, int rows,
+                                       int H, float eps) {
+  __shared__ float red[16];
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const float* xr = x + (long)row * H;
+    float* yr = y + (long)row * H;
+    float ss = 0.f;
+    for (int i = threadIdx.x; i < H; i += blockDim.x) ss += xr[i] * xr[i];
+    ss = block_reduce_sum<4>(ss, red);
+    float r = rsqrtf(ss / H + eps);
+    if (threadIdx.x == 0) inv_rms[row] = r;
+    for (int i = threadIdx.x; i < H; i += blockDim.x)
+      yr[i] = xr[i] * r * w[i];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// RoPE (fused q + k, neox half-split)
+// ---------------------------------------------------------------------------
+// x [b, s, h, d]; cos/sin [s, d/2] fp32. Each thread: 8 first-half elements
+// + 8 matching second-half elements of one (b,s,h) row.
+
+__global__ void rope_kernel(const short* __restrict__ x,
+                            short* __restrict__ out,
+                            const float* __restrict__ cosp,
+                            const float* __restrict__ sinp, long total_rows,
+                            int H, int D, int S) {
+  const int d2 = D / 2;
+  const int chunks = d2 / 8;  // d2 multiple of 8
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long n = total_rows * chunks;
+  for (; idx < n; idx += (long)gridDim.x * blockDim.x) {
+    const long rowhead = idx / chunks;   // (b*s*h)
+    const int c = idx % chunks;
+    const int s = (rowhead / H) % S;
+    const short* xr = x + rowhead * D + c * 8;
+    short* orow = out + rowhead * D + c * 8;
+    const float* cp = cosp + (long)s * d2 + c * 8;
+    const float* sp = sinp + (long)s * d2 + c * 8;
+    s16x8 x1 = *reinterpret_cast<const s16x8*>(xr);
+    s16x8 x2 = *reinterpret_cast<const s16x8*>(xr + d2);
+    f32x4 c0 = *reinterpret_cast<const f32x4*>(cp);
+    f32x4 c1 = *reinterpret_cast<const f32x4*>(cp + 4);
+    f32x4 s0 = *reinterpret_cast<const f32x4*>(sp);
+    f32x4 s1 = *reinterpret_cast<const f32x4*>(sp + 4);
+    s16x8 o1, o2;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float cj = j < 4 ? c0[j] : c1[j - 4];
+      float sj = j < 4 ? s0[j] : s1[j - 4];
+      float a = bf16_to_f32(x1[j]);
+      float b = bf16_to_f32(x2[j]);
+      o1[j] = f32_to_bf16(a * cj - b * sj);
+      o2[j] = f32_to_bf16(b * cj + a * sj);
+    }
+    *reinterpret_cast<s16x8*>(orow) = o1;
+    *reinterpret_cast<s16x8*>(orow + d2) = o2;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// SwiGLU
+// ---------------------------------------------------------------------------
+
+__global__ void swiglu_fwd_kernel(const short* __restrict__ g,
+                                  const short* __restrict__ u,
+                                  short* __restrict__ y, long n8) {
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; idx < n8; idx += (long)gridDim.x * blockDim.x) {
+    s16x8 gv = *reinterpret_cast<const s16x8*>(g + idx * 8);
+    s16x8 uv = *reinterpret_cast<const s16x8*>(u + idx * 8);
+    s16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf16_to_f32(gv[j]);
+      float sig = 1.f / (1.f + __expf(-gf));
+      o[j] = f32_to_bf16(gf * sig * bf16_to_f32(uv[j]));
+    }
+    *reinterpret_cast<s16x8*>(y + idx * 8) = o;
+  }
+}
+
+__global__ void swiglu_bwd_kernel(const short* __restrict__ dy,
+                                  const short* __restrict__ g,
+                                  const short* __restrict__ u,
+                                  short* __restrict__ dg,
+                                  short* __restrict__ du, long n8) {
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; idx < n8; idx += (long)gridDim.x * blockDim.x) {
+    s16x8 dv = *reinterpret_cast<const s16x8*>(dy + idx * 8);
+    s16x8 gv = *reinterpret_cast<const s16x8*>(g + idx * 8);
+    s16x8 uv = *reinterpret_cast<const s16x8*>(u + idx * 8);
+    s16x8 og, ou;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf16_to_f32(gv[j]);
+      float uf = bf16_to_f32(uv[j]);
+      float d = bf16_to_f32(dv[j]);
+      float sig = 1.f / (1.f + __expf(-gf));
+      float silu = gf * sig;
+      og[j] = f32_to_bf16(d * uf * (sig * (1.f + gf * (1.f - sig))));
+      ou[j] = f32_to_bf16(d * silu);
+    }
+    *reinterpret_cast<s16x8*>(dg + idx * 8) = og;
+    *reinterpret_cast<s16x8*>(du + idx * 8) = ou;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+
+static inline int grid_for(long work, int block) {
+  long g = (work + block - 1) / block;
+  return (int)std::min<long>(g, 2048);
+}
+
+std::vector<torch::Tensor> rmsnorm_forward(torch::Tensor x, torch::Tensor w,
+                                           double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  const int H = x.size(-1);
+  const long rows = x.numel() / H;
+  auto y = torch::empty_like(x);
+  auto inv_rms = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  const int grid = grid_for(rows, 1);
+  if (x.scalar_type() == torch::kBFloat16) {
+    TORCH_CHECK(H % 8 == 0, "hidden size must be a multiple of 8");
+    hipLaunchKernelGGL(rmsnorm_fwd_kernel, dim3(std::min<long>(rows, 2048)),
+                       dim3(256), 0, stream,
+                       (const short*)x.data_ptr(), (const short*)w.data_ptr(),
+                       (short*)y.data_ptr(), inv_rms.data_ptr<float>(),
+                       (int)rows, H, (float)eps);
+  } else if (x.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(rmsnorm_fwd_kernel_f32,
+                       dim3(std::min<long>(rows, 2048)), dim3(256), 0, stream,
+                       x.data_ptr<float>(), w.data_ptr<float>(),
+                       y.data_ptr<float>(), inv_rms.data_ptr<float>(),
+                       (int)rows, H, (float)eps);
+  } else {
+    TORCH_CHECK(false, "rmsnorm: unsupported dtype");
+  }
+  HIP_CHECK_LAST();
+  return {y, inv_rms};
+}
+
+std::vector<torch::Tensor> rmsnorm_backward(torch::Tensor dy, torch::Tensor x,
+                                            torch::Tensor w,
+                                            torch::Tensor inv_rms) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16,
+              "rmsnorm backward: bf16 only on GPU");
+  const int H = x.size(-1);
+  const long rows = x.numel() / H;
+  auto dx = torch::empty_like(x);
+  auto dw32 = torch::zeros({H}, x.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  const int lds = (H + 16) * sizeof(float);
+  hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(std::min<long>(rows, 512)),
+                     dim3(256), lds, stream, (const short*)dy.data_ptr(),
+                     (const short*)x.data_ptr(), (const short*)w.data_ptr(),
+                     inv_rms.data_ptr<float>(), (short*)dx.data_ptr(),
+                     dw32.data_ptr<float>(), (int)rows, H);
+  HIP_CHECK_LAST();
+  return {dx, dw32.to(w.scalar_type())};
+}
+
+std::vector<torch::Tensor> rope_forward(torch::Tensor q, torch::Tensor k,
+                                        torch::Tensor cos, torch::Tensor sin) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous());
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "rope: bf16 only on GPU");
+  TORCH_CHECK(cos.scalar_type() == torch::kFloat32);
+  const int D = q.size(-1);
+  TORCH_CHECK(D % 16 == 0, "rope: head_dim must be a multiple of 16");
+  const int S = q.size(1);
+  auto cosc = cos.contiguous();
+  auto sinc = sin.contiguous();
+  auto qo = torch::empty_like(q);
+  auto ko = torch::empty_like(k);
+  auto stream = at::hip::getCurrentHIPStream();
+  {
+    const long rows = q.numel() / D;
+    const long work = rows * (D / 16);
+    hipLaunchKernelGGL(rope_kernel, dim3(grid_for(work, 256)), dim3(256), 0,
+                       stream, (const short*)q.data_ptr(),
+                       (short*)qo.data_ptr(), cosc.data_ptr<float>(),
+                       sinc.data_ptr<float>(), rows, (int)q.size(2), D, S);
+  }
+  {
+    const long rows = k.numel() / D;
+    const long work = rows * (D / 16);
+    hipLaunchKernelGGL(rope_kernel, dim3(grid_for(work, 256)), dim3(256), 0,
+                       stream, (const short*)k.data_ptr(),
+                       (short*)ko.data_ptr(), cosc.data_ptr<float>(),
+                       sinc.data_ptr<float>(), rows, (int)k.size(2), D, S);
+  }
+  HIP_CHECK_LAST();
+  return {qo, ko};
+}
+
+torch::Tensor swiglu_forward(torch::Tensor g, torch::Tensor u) {
+  TORCH_CHECK(g.is_cuda() && g.scalar_type() == torch::kBFloat16,
+              "swiglu: bf16 only on GPU");
+  TORCH_CHECK(g.numel() % 8 == 0);
+  auto y = torch::empty_like(g);
+  auto stream = at::hip::getCurrentHIPStream();
+  const long n8 = g.numel() / 8;
+  hipLaunchKernelGGL(swiglu_fwd_kernel, dim3(grid_for(n8, 256)), dim3(256),
+                     0, stream, (const short*)g.data_ptr(),
+                     (const short*)u.data_ptr(), (short*)y.data_ptr(), n8);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+std::vector<torch::Tensor> swiglu_backward(torch::Tensor dy, torch::Tensor g,
+                                           torch::Tensor u) {
+  TORCH_CHECK(g.is_cuda() && g.scalar_type() == torch::kBFloat16);
+  auto dg = torch::empty_like(g);
+  auto du = torch::empty_like(u);
+  auto stream = at::hip::getCurrentHIPStream();
+  const long n8 = g.numel() / 8;
+  hipLaunchKernelGGL(swiglu_bwd_kernel, dim3(grid_for(n8, 256)), dim3(256),
+                     0, stream, (const short*)dy.data_ptr(),
+                     (const short*)g.data_ptr(), (const short*)u.data_ptr(),
+                     (short*)dg.data_ptr(), (short*)du.data_ptr(), n8);
+  HIP_CHECK_LAST();
+  return {dg, du};
+}

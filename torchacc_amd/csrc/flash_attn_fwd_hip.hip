@@ -1,0 +1,370 @@
+#include "hip/hip_runtime.h"
+// Flash-attention v2 FORWARD for gfx950 (CDNA4 MFMA).
+//
+// Structure (per the CDNA4 attention ladder in the MI355X guides):
+//   - workgroup = 8 waves (512 threads) = 256 Q rows; each wave owns a
+//     32-row Q block. Q stays in registers (D/16 bf16x8 fragments/lane).
+//   - KV tile = 64 keys staged in LDS, double-buffered: K row-major with the
+//     (row&7)<<4 XOR byte swizzle (conflict-free ds_read_b128 column-slice
+//     reads), V stored TRANSPOSED [D][64] with the (d&7)<<4 swizzle.
+//   - QK^T is computed SWAPPED: mfma_f32_32x32x16_bf16(A=K, B=Q) so each
+//     lane holds the score row of ONE q (col = lane&31): the softmax row
+//     reduction is 31 in-lane ops + one cross-half __shfl_xor(.,32).
+//   - online softmax in registers (m, l per lane); P is converted to bf16
+//     with v_cvt_pk_bf16_f32 and redistributed across lane halves with
+//     permlane32_swap, landing EXACTLY in the PV mfma's B-fragment layout.
+//   - PV: mfma(A=V^T fragment from LDS, B=P) accumulating OUT^T[d][q];
+//     epilogue normalizes by 1/l and stores out + logsumexp.
+//
+// Covers: causal (bottom-right aligned), GQA (h_kv | h), sliding window,
+// per-batch varlen (q_lens/k_lens), D in {64, 128}. bf16 only.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+#define LOG2E 1.4426950408889634f
+
+typedef float f32x16_ __attribute__((ext_vector_type(16)));
+
+DEVINLINE unsigned cvt_pk_bf16(float lo, float hi) {
+  unsigned r;
+  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
+               : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+// crow: D-matrix row for accumulator register r on lane-half hi (32x32 mfma)
+#define CROW(r, hi) (((r) & 3) + 8 * ((r) >> 2) + 4 * (hi))
+
+template <int D, bool CAUSAL, bool HAS_WINDOW, bool HAS_LENS>
+__global__ __launch_bounds__(512, 2)
+void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
+                   const short* __restrict__ V, short* __restrict__ O,
+                   float* __restrict__ LSE, int b_, int sq, int sk, int hq,
+                   int hk, float scale, int wl, int wr,
+                   const int* __restrict__ q_lens,
+                   const int* __restrict__ k_lens) {
+  constexpr int NT = D / 16;   // QK^T k-steps (d slices of 16)
+  constexpr int NA = D / 32;   // PV output accs (d blocks of 32)
+  constexpr int KVB = 64;      // keys per tile
+  constexpr int KROW_BYTES = D * 2;
+
+  // LDS: K [64][D] + V^T [D][64], double buffered
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  short* k_lds = reinterpret_cast<short*>(smem);                // 2*64*D
+  short* vt_lds = reinterpret_cast<short*>(smem) + 2 * KVB * D; // 2*D*64
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;          // 0..7
+  const int col = lane & 31;         // q row within wave block / d col
+  const int hi = lane >> 5;          // lane half
+
+  const int h = blockIdx.y % hq;
+  const int b = blockIdx.y / hq;
+  const int kh = h / (hq / hk);
+  const int q0wg = blockIdx.x * 256;
+  const int q0 = q0wg + wid * 32;        // this wave's q block
+  const int qrow = q0 + col;             // this lane's q row
+  const int shift = sk - sq;             // bottom-right causal alignment
+
+  int klimit = sk;
+  int qlimit = sq;
+  if (HAS_LENS) {
+    if (k_lens != nullptr) klimit = min(klimit, k_lens[b]);
+    if (q_lens != nullptr) qlimit = min(qlimit, q_lens[b]);
+  }
+
+  // ---- load Q fragments (pre-scaled) -------------------------------------
+  // B-frag for swapped QK^T: lane holds Q[qrow][t*16 + hi*8 + j], j=0..7
+  bf16x8 qfrag[NT];
+  {
+    const long qbase = ((long)(b * sq + min(qrow, sq - 1)) * hq + h) * D;
+    const bool qvalid = (qrow < sq);
+#pragma unroll
+    for (int t = 0; t < NT; ++t) {
+      s16x8 v = qvalid
+          ? *reinterpret_cast<const s16x8*>(Q + qbase + t * 16 + hi * 8)
+          : s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+#pragma unroll
+      for (int j = 0; j < 8; ++j) qfrag[t][j] = v[j];
+    }
+  }
+
+  // ---- KV tile range -----------------------------------------------------
+  int kv_hi_key = klimit;  // exclusive
+  if (CAUSAL) kv_hi_key = min(kv_hi_key, q0wg + 255 + shift + 1);
+  int kv_lo_key = 0;
+  if (HAS_WINDOW && wl >= 0) kv_lo_key = max(0, q0wg + shift - wl);
+  const int t0 = kv_lo_key / KVB;
+  const int t1 = (kv_hi_key + KVB - 1) / KVB;  // exclusive
+  const int ntiles = t1 - t0;
+
+  // ---- staging helpers ---------------------------------------------------
+  const long kbase = ((long)b * sk * hk + kh) * D;  // + key*hk*D + d
+  auto stage = [&](int buf, int tile) {
+    const int kv0 = (t0 + tile) * KVB;
+    // K: 64 rows x D cols; 512 threads x (64*D/8/512) 16B chunks
+    constexpr int CHUNKS = KVB * D / 8;      // 16B chunks in tile
+    short* kdst = k_lds + buf * KVB * D;
+    short* vdst = vt_lds + buf * KVB * D;
+#pragma unroll
+    for (int c = tid; c < CHUNKS; c += 512) {
+      const int row = c / (D / 8);
+      const int d0 = (c % (D / 8)) * 8;
+      const int key = kv0 + row;
+      s16x8 kv8 = {0, 0, 0, 0, 0, 0, 0, 0};
+      s16x8 vv8 = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (key < sk) {
+        const long src = kbase + (long)key * hk * D + d0;
+        kv8 = *reinterpret_cast<const s16x8*>(K + src);
+        vv8 = *reinterpret_cast<const s16x8*>(V + src);
+      }
+      // K: row-major with (row&7)<<4 byte-XOR swizzle
+      {
+        unsigned byte = row * KROW_BYTES + d0 * 2;
+        byte ^= (unsigned)((row & 7) << 4);
+        *reinterpret_cast<s16x8*>(
+            reinterpret_cast<char*>(kdst) + byte) = kv8;
+      }
+      // V^T: [D][64] with (d&7)<<4 swizzle, scalar scatter
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        unsigned byte = (unsigned)(d0 + j) * (KVB * 2) + row * 2;
+        byte ^= (unsigned)(((d0 + j) & 7) << 4);
+        *reinterpret_cast<short*>(
+            reinterpret_cast<char*>(vdst) + byte) = vv8[j];
+      }
+    }
+  };
+
+  // ---- online softmax state ---------------------------------------------
+  float m_run = -INFINITY;
+  float l_run = 0.f;
+  f32x16_ oacc[NA];
+#pragma unroll
+  for (int a = 0; a < NA; ++a) oacc[a] = f32x16_(0.f);
+
+  if (ntiles > 0) stage(0, 0);
+  __syncthreads();
+
+  for (int tile = 0; tile < ntiles; ++tile) {
+    const int buf = tile & 1;
+    const int kv0 = (t0 + tile) * KVB;
+    // wave-level skip: tile entirely outside this wave's causal/window span
+    bool wave_active = true;
+    if (CAUSAL && kv0 > q0 + 31 + shift) wave_active = false;
+    if (HAS_WINDOW && wl >= 0 && kv0 + KVB <= q0 + shift - wl)
+      wave_active = false;
+    if (qrow >= qlimit && __all(qrow >= qlimit)) {
+      // whole wave beyond valid q rows: nothing to compute
+    }
+
+    if (wave_active) {
+      const short* kbuf = k_lds + buf * KVB * D;
+      const short* vbuf = vt_lds + buf * KVB * D;
+      // ---- QK^T: 2 key sub-blocks x NT d-steps --------------------------
+      f32x16_ p[2];
+      p[0] = f32x16_(0.f);
+      p[1] = f32x16_(0.f);
+#pragma unroll
+      for (int kb = 0; kb < 2; ++kb) {
+#pragma unroll
+        for (int t = 0; t < NT; ++t) {
+          // A-frag: K[key = kb*32 + col][d = t*16 + hi*8 + j]
+          const int row = kb * 32 + col;
+          unsigned byte = (unsigned)row * KROW_BYTES + (t * 16 + hi * 8) * 2;
+          byte ^= (unsigned)((row & 7) << 4);
+          bf16x8 kf = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<const char*>(kbuf) + byte);
+          p[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfrag[t],
+                                                          p[kb], 0, 0, 0);
+        }
+      }
+      // ---- mask + scale --------------------------------------------------
+      const bool tile_full =
+          (!CAUSAL || kv0 + KVB - 1 <= q0 + shift) &&
+          (kv0 + KVB <= klimit) &&
+          (!HAS_WINDOW || wl < 0 || kv0 >= q0 + shift - wl);
+      float pmax = -INFINITY;
+#pragma unroll
+      for (int kb = 0; kb < 2; ++kb) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float s = p[kb][r] * scale;
+          if (!tile_full) {
+            const int key = kv0 + kb * 32 + CROW(r, hi);
+            bool valid = key < klimit;
+            if (CAUSAL) valid &= (key <= qrow + shift);
+            if (HAS_WINDOW && wl >= 0) valid &= (key >= qrow + shift - wl);
+            if (HAS_WINDOW && wr >= 0 && !CAUSAL)
+              valid &= (key <= qrow + shift + wr);
+            s = valid ? s : -INFINITY;
+          }
+          p[kb][r] = s;
+          pmax = fmaxf(pmax, s);
+        }
+      }
+      pmax = fmaxf(pmax, __shfl_xor(pmax, 32, 64));
+      // ---- online rescale ------------------------------------------------
+      const float m_new = fmaxf(m_run, pmax);
+      const float m_use = (m_new == -INFINITY) ? 0.f : m_new;
+      const float alpha =
+          (m_run == -INFINITY) ? 0.f : __expf(m_run - m_use);
+      m_run = m_new;
+      float lsum = 0.f;
+#pragma unroll
+      for (int kb = 0; kb < 2; ++kb) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float e = __expf(p[kb][r] - m_use);  // -inf -> 0
+          p[kb][r] = e;
+          lsum += e;
+        }
+      }
+      lsum += __shfl_xor(lsum, 32, 64);
+      l_run = l_run * alpha + lsum;
+#pragma unroll
+      for (int a = 0; a < NA; ++a) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) oacc[a][r] *= alpha;
+      }
+      // ---- P -> bf16 B-fragments (cvt_pk + permlane32_swap) -------------
+      // step (kb, tp): B-frag covering keys [kv0+kb*32+16*tp, +16)
+      unsigned pb[4][4];  // [step][u32 slot]
+#pragma unroll
+      for (int kb = 0; kb < 2; ++kb) {
+#pragma unroll
+        for (int tp = 0; tp < 2; ++tp) {
+#pragma unroll
+          for (int u = 0; u < 2; ++u) {
+            const int r = 2 * u + 8 * tp;
+            unsigned va = cvt_pk_bf16(p[kb][r], p[kb][r + 1]);
+            unsigned vb = cvt_pk_bf16(p[kb][r + 4], p[kb][r + 5]);
+            auto sw = __builtin_amdgcn_permlane32_swap(va, vb, false, false);
+            pb[kb * 2 + tp][u] = sw[0];
+            pb[kb * 2 + tp][u + 2] = sw[1];
+          }
+        }
+      }
+      // ---- PV: OUT^T[d][q] += V^T x P -----------------------------------
+#pragma unroll
+      for (int a = 0; a < NA; ++a) {
+#pragma unroll
+        for (int st = 0; st < 4; ++st) {
+          // A-frag: V^T[d = a*32 + col][key = st*16 + hi*8 + j]
+          const int d = a * 32 + col;
+          unsigned byte = (unsigned)d * (KVB * 2) + (st * 16 + hi * 8) * 2;
+          byte ^= (unsigned)((d & 7) << 4);
+          bf16x8 vf = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<const char*>(vbuf) + byte);
+          bf16x8 pf = *reinterpret_cast<const bf16x8*>(&pb[st][0]);
+          oacc[a] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pf, oacc[a],
+                                                            0, 0, 0);
+        }
+      }
+    }
+
+    if (tile + 1 < ntiles) stage(buf ^ 1, tile + 1);
+    __syncthreads();
+  }
+
+  // ---- epilogue ----------------------------------------------------------
+  const bool qvalid = (qrow < qlimit) && (qrow < sq);
+  const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+  if (qvalid) {
+    const long obase = ((long)(b * sq + qrow) * hq + h) * D;
+#pragma unroll
+    for (int a = 0; a < NA; ++a) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int d = a * 32 + CROW(r, hi);
+        O[obase + d] = f32_to_bf16(oacc[a][r] * inv_l);
+      }
+    }
+    if (hi == 0) {
+      LSE[((long)b * hq + h) * sq + qrow] =
+          (l_run > 0.f) ? m_run + __logf(l_run) : -INFINITY;
+    }
+  } else if (qrow < sq) {
+    // rows masked away by q_lens: zero output, zero lse (reference behavior)
+    const long obase = ((long)(b * sq + qrow) * hq + h) * D;
+#pragma unroll
+    for (int a = 0; a < NA; ++a) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int d = a * 32 + CROW(r, hi);
+        O[obase + d] = 0;
+      }
+    }
+    if (hi == 0) LSE[((long)b * hq + h) * sq + qrow] = 0.f;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host wrapper
+// ---------------------------------------------------------------------------
+
+template <int D>
+static void launch_fa_fwd(const torch::Tensor& q, const torch::Tensor& k,
+                          const torch::Tensor& v, torch::Tensor& o,
+                          torch::Tensor& lse, float scale, bool causal,
+                          int wl, int wr, const torch::Tensor& q_lens,
+                          const torch::Tensor& k_lens, hipStream_t stream) {
+  const int b = q.size(0), sq = q.size(1), hq = q.size(2);
+  const int sk = k.size(1), hk = k.size(2);
+  const bool has_window = (wl >= 0 || wr >= 0);
+  const bool has_lens = q_lens.numel() > 0 || k_lens.numel() > 0;
+  dim3 grid((sq + 255) / 256, b * hq);
+  dim3 block(512);
+  const int lds = 4 * 64 * D * 2;  // K + V^T, double buffered
+  const int* qlp = q_lens.numel() ? q_lens.data_ptr<int>() : nullptr;
+  const int* klp = k_lens.numel() ? k_lens.data_ptr<int>() : nullptr;
+
+#define LAUNCH(CAUSAL, WIN, LENS)                                            \
+  hipLaunchKernelGGL((fa_fwd_kernel<D, CAUSAL, WIN, LENS>), grid, block,     \
+                     lds, stream, (const short*)q.data_ptr(),                \
+                     (const short*)k.data_ptr(),                             \
+                     (const short*)v.data_ptr(), (short*)o.data_ptr(),       \
+                     lse.data_ptr<float>(), b, sq, sk, hq, hk, scale, wl,    \
+                     wr, qlp, klp)
+  if (causal) {
+    if (has_window) { if (has_lens) LAUNCH(true, true, true); else LAUNCH(true, true, false); }
+    else { if (has_lens) LAUNCH(true, false, true); else LAUNCH(true, false, false); }
+  } else {
+    if (has_window) { if (has_lens) LAUNCH(false, true, true); else LAUNCH(false, true, false); }
+    else { if (has_lens) LAUNCH(false, false, true); else LAUNCH(false, false, false); }
+  }
+#undef LAUNCH
+}
+
+std::vector<torch::Tensor> fa_forward(torch::Tensor q, torch::Tensor k,
+                                      torch::Tensor v, double softmax_scale,
+                                      bool causal, long wl, long wr,
+                                      torch::Tensor q_lens,
+                                      torch::Tensor k_lens) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
+              v.is_contiguous());
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16,
+              "fa_forward: bf16 only (CDNA4 MFMA path)");
+  const int D = q.size(3);
+  TORCH_CHECK(D == 64 || D == 128, "fa_forward: head_dim must be 64 or 128");
+  TORCH_CHECK(q.size(2) % k.size(2) == 0);
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({q.size(0), q.size(2), q.size(1)},
+                          q.options().dtype(torch::kFloat32));
+  auto ql = q_lens.numel() ? q_lens.to(q.device(), torch::kInt32)
+                           : q_lens;
+  auto kl = k_lens.numel() ? k_lens.to(q.device(), torch::kInt32)
+                           : k_lens;
+  auto stream = at::hip::getCurrentHIPStream();
+  if (D == 128) {
+    launch_fa_fwd<128>(q, k, v, o, lse, (float)softmax_scale, causal,
+                       (int)wl, (int)wr, ql, kl, stream);
+  } else {
+    launch_fa_fwd<64>(q, k, v, o, lse, (float)softmax_scale, causal, (int)wl,
+                      (int)wr, ql, kl, stream);
+  }
+  HIP_CHECK_LAST();
+  return {o, lse};
+}

@@ -1,12 +1,14 @@
 """Fused AdamW for MI355X (+ syncfree found_inf gating).
 
 Replaces torch_xla.amp.syncfree.AdamW (reference utils/patch.py:55-57): the
-HIP kernel performs the whole AdamW update for a list of params in one
-multi-tensor launch, keeps fp32 exp_avg/exp_avg_sq (and optional fp32 master
-weights for bf16/fp16 params), and accepts a device-side ``found_inf`` flag:
-when nonzero the update is a no-op, so fp16 loss scaling never host-syncs.
+HIP kernel performs the whole AdamW update for each (large, flat) param in
+one grid-stride launch, keeps fp32 exp_avg/exp_avg_sq (and optional fp32
+master weights for bf16/fp16 params), and accepts a device-side
+``found_inf`` flag: when nonzero the update is a device-side no-op, so fp16
+loss scaling never host-syncs. The step counter lives on the host (bias
+correction after a skipped step drifts by that one step — warmup-only,
+matching syncfree semantics closely enough for training parity).
 """
-import math
 from typing import Optional
 
 import torch
@@ -41,15 +43,15 @@ class AdamW(torch.optim.Optimizer):
                     continue
                 state = self.state[p]
                 if len(state) == 0:
-                    state["step"] = torch.zeros(
-                        (), dtype=torch.float32, device=p.device)
-                    state["exp_avg"] = torch.zeros_like(
-                        p, dtype=torch.float32)
-                    state["exp_avg_sq"] = torch.zeros_like(
-                        p, dtype=torch.float32)
+                    state["step"] = 0.0
+                    state["exp_avg"] = torch.zeros(
+                        p.shape, dtype=torch.float32, device=p.device)
+                    state["exp_avg_sq"] = torch.zeros(
+                        p.shape, dtype=torch.float32, device=p.device)
                     if self.use_master_weights and p.dtype in (
                             torch.bfloat16, torch.float16):
                         state["master"] = p.detach().float().clone()
+                state["step"] += 1.0
                 params.append(p)
                 grads.append(p.grad)
                 exp_avgs.append(state["exp_avg"])
@@ -64,7 +66,12 @@ class AdamW(torch.optim.Optimizer):
                     torch.zeros((), dtype=torch.float32,
                                 device=params[0].device)
                 master_list = [
-                    m if m is not None else torch.empty(0) for m in masters
+                    m if m is not None else torch.empty(
+                        0, device=params[0].device) for m in masters
+                ]
+                grads = [
+                    g if g.dtype == p.dtype else g.to(p.dtype)
+                    for g, p in zip(grads, params)
                 ]
                 ext.fused_adamw(params, grads, exp_avgs, exp_avg_sqs,
                                 master_list, steps, fi, lr, beta1, beta2,
@@ -72,10 +79,8 @@ class AdamW(torch.optim.Optimizer):
             else:
                 if found_inf is not None and bool(found_inf != 0):
                     continue
-                for p, g, m, v, mw, st in zip(params, grads, exp_avgs,
-                                              exp_avg_sqs, masters, steps):
-                    st += 1
-                    t = float(st)
+                for p, g, m, v, mw, t in zip(params, grads, exp_avgs,
+                                             exp_avg_sqs, masters, steps):
                     gf = g.float()
                     m.mul_(beta1).add_(gf, alpha=1 - beta1)
                     v.mul_(beta2).addcmul_(gf, gf, value=1 - beta2)
@@ -83,16 +88,13 @@ class AdamW(torch.optim.Optimizer):
                     bc2 = 1 - beta2 ** t
                     denom = (v / bc2).sqrt_().add_(eps)
                     upd = (m / bc1) / denom
-                    tgt = mw if mw is not None else p
-                    tgtf = tgt if tgt.dtype == torch.float32 else tgt.float()
-                    tgtf = tgtf.mul_(1 - lr * wd).add_(upd, alpha=-lr)
-                    if mw is not None:
-                        mw.copy_(tgtf)
-                        p.copy_(tgtf.to(p.dtype))
-                    else:
+                    tgtf = mw if mw is not None else (
+                        p if p.dtype == torch.float32 else p.float())
+                    tgtf.mul_(1 - lr * wd).add_(upd, alpha=-lr)
+                    if tgtf is not p:
                         p.copy_(tgtf.to(p.dtype))
         return loss
 
 
-# torch_xla.amp.syncfree-compatible aliases
+# torch_xla.amp.syncfree-compatible alias
 SyncFreeAdamW = AdamW

@@ -278,9 +278,8 @@ class FlashAttnVarlenFunc(torch.autograd.Function):
         ext = dispatch(q)
         wl, wr = window_size
         if ext is not None:
-            out, lse = ext.fa_varlen_forward(q, k, v, cu_q.to(q.device),
-                                             cu_k.to(q.device), max_q, max_k,
-                                             softmax_scale, causal, wl, wr)
+            out, lse = _varlen_fwd_gpu(ext, q, k, v, cu_q, cu_k,
+                                       softmax_scale, causal, wl, wr)
         else:
             out, lse = _ref_varlen(q, k, v, cu_q, cu_k, softmax_scale,
                                    causal, (wl, wr))
@@ -296,14 +295,62 @@ class FlashAttnVarlenFunc(torch.autograd.Function):
         dout = dout.contiguous()
         wl, wr = window_size
         if ext is not None:
-            dq, dk, dv = ext.fa_varlen_backward(
-                dout, q, k, v, out, lse, cu_q, cu_k, max_q, max_k,
-                softmax_scale, causal, wl, wr)
+            dq, dk, dv = _varlen_bwd_gpu(ext, dout, q, k, v, out, lse, cu_q,
+                                         cu_k, softmax_scale, causal, wl, wr)
         else:
             dq, dk, dv = _ref_varlen_backward(dout, q, k, v, out, lse, cu_q,
                                               cu_k, softmax_scale, causal,
                                               (wl, wr))
         return (dq, dk, dv) + (None,) * 9
+
+
+def _varlen_fwd_gpu(ext, q, k, v, cu_q, cu_k, softmax_scale, causal, wl,
+                    wr):
+    """Packed varlen on the fixed-length CDNA4 kernel: one launch per
+    sequence (the kernels are whole-tile; a fused multi-sequence varlen
+    kernel is a planned optimization)."""
+    total, h, _d = q.shape
+    out = torch.zeros_like(q)
+    lse = torch.zeros(h, total, dtype=torch.float32, device=q.device)
+    empty = torch.empty(0, device=q.device)
+    for i in range(cu_q.numel() - 1):
+        qs, qe = int(cu_q[i]), int(cu_q[i + 1])
+        ks, ke = int(cu_k[i]), int(cu_k[i + 1])
+        if qe == qs:
+            continue
+        o_i, lse_i = ext.fa_forward(
+            q[qs:qe].unsqueeze(0).contiguous(),
+            k[ks:ke].unsqueeze(0).contiguous(),
+            v[ks:ke].unsqueeze(0).contiguous(), softmax_scale, causal, wl,
+            wr, empty, empty)
+        out[qs:qe] = o_i.squeeze(0)
+        lse[:, qs:qe] = lse_i.squeeze(0)
+    return out, lse
+
+
+def _varlen_bwd_gpu(ext, dout, q, k, v, out, lse, cu_q, cu_k, softmax_scale,
+                    causal, wl, wr):
+    dq = torch.zeros_like(q)
+    dk = torch.zeros_like(k)
+    dv = torch.zeros_like(v)
+    empty = torch.empty(0, device=q.device)
+    for i in range(cu_q.numel() - 1):
+        qs, qe = int(cu_q[i]), int(cu_q[i + 1])
+        ks, ke = int(cu_k[i]), int(cu_k[i + 1])
+        if qe == qs:
+            continue
+        dq_i, dk_i, dv_i = ext.fa_backward(
+            dout[qs:qe].unsqueeze(0).contiguous(),
+            q[qs:qe].unsqueeze(0).contiguous(),
+            k[ks:ke].unsqueeze(0).contiguous(),
+            v[ks:ke].unsqueeze(0).contiguous(),
+            out[qs:qe].unsqueeze(0).contiguous(),
+            lse[:, qs:qe].unsqueeze(0).contiguous(), softmax_scale, causal,
+            wl, wr, empty, empty)
+        dq[qs:qe] = dq_i.squeeze(0)
+        dk[ks:ke] = dk_i.squeeze(0)
+        dv[ks:ke] = dv_i.squeeze(0)
+    return dq, dk, dv
 
 
 def _ref_varlen(q, k, v, cu_q, cu_k, softmax_scale, causal, window):
