@@ -90,7 +90,12 @@ def main():
         cfg.memory.gc = True
         cfg.memory.gc_cls = {"LlamaDecoderLayer"}
 
-    model, mcfg = build_model(args, cfg)
+    if on_gpu:
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", 0)))
+        with torch.device("cuda"):
+            model, mcfg = build_model(args, cfg)
+    else:
+        model, mcfg = build_model(args, cfg)
     model = ta.accelerate(model, config=cfg)
     opt = ta.ops.AdamW(model.parameters(), lr=1e-4, weight_decay=0.0)
 

@@ -182,10 +182,15 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
         }
       }
       // ---- mask + scale --------------------------------------------------
+      // "full" = no per-element masking needed for ANY q row of this wave:
+      // causal bound uses the wave's SMALLEST q (q0), window-left bound its
+      // LARGEST q (q0+31), window-right its smallest.
       const bool tile_full =
           (!CAUSAL || kv0 + KVB - 1 <= q0 + shift) &&
           (kv0 + KVB <= klimit) &&
-          (!HAS_WINDOW || wl < 0 || kv0 >= q0 + shift - wl);
+          (!HAS_WINDOW || wl < 0 || kv0 >= q0 + 31 + shift - wl) &&
+          (!HAS_WINDOW || wr < 0 || CAUSAL ||
+           kv0 + KVB - 1 <= q0 + shift + wr);
       float pmax = -INFINITY;
 #pragma unroll
       for (int kb = 0; kb < 2; ++kb) {

@@ -46,6 +46,8 @@ class _RoPE(torch.autograd.Function):
         ext = dispatch(q)
         q = q.contiguous()
         k = k.contiguous()
+        cos = cos.float()
+        sin = sin.float()
         if ext is not None:
             qo, ko = ext.rope_forward(q, k, cos, sin)
         else:
@@ -56,6 +58,8 @@ class _RoPE(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dq, dk):
         cos, sin = ctx.saved_tensors
+        cos = cos.float()
+        sin = sin.float()
         ext = dispatch(dq)
         dq = dq.contiguous()
         dk = dk.contiguous()
