@@ -105,9 +105,17 @@ def main():
     is_cp = args.mode in ("ulysses", "ring", "2d")
     local_seq = seq // world if is_cp else seq
     torch.manual_seed(5678 + rank if not is_cp else 5678)
-    ids = torch.randint(0, mcfg.vocab_size, (bs, local_seq), device=device)
+    # a few distinct synthetic batches, cycled: keeps the loss honest (no
+    # single-batch memorization) without growing device memory
+    batches = [
+        torch.randint(0, mcfg.vocab_size, (bs, local_seq), device=device)
+        for _ in range(4)
+    ]
+    it = [0]
 
     def step_fixed():
+        ids = batches[it[0] % len(batches)]
+        it[0] += 1
         # labels are shifted inside the model (predict t+1 from t)
         loss = model(ids, labels=ids)
         loss.backward()

@@ -48,7 +48,11 @@ __global__ void ce_fwd_kernel(const short* __restrict__ logits,
       float mo = __shfl_xor(m, off, 64);
       float so = __shfl_xor(s, off, 64);
       float mn = fmaxf(m, mo);
-      s = s * __expf(m - mn) + so * __expf(mo - mn);
+      // guard -inf slices (threads whose vocab slice was empty):
+      // exp(-inf - -inf) would be NaN
+      float wa = (m == -INFINITY) ? 0.f : __expf(m - mn);
+      float wb = (mo == -INFINITY) ? 0.f : __expf(mo - mn);
+      s = s * wa + so * wb;
       m = mn;
     }
     __shared__ float mred[8], sred[8];
@@ -58,7 +62,9 @@ __global__ void ce_fwd_kernel(const short* __restrict__ logits,
       float mm = mred[0], ss2 = sred[0];
       for (int wv = 1; wv < blockDim.x / WAVE; ++wv) {
         float mn = fmaxf(mm, mred[wv]);
-        ss2 = ss2 * __expf(mm - mn) + sred[wv] * __expf(mred[wv] - mn);
+        float wa = (mm == -INFINITY) ? 0.f : __expf(mm - mn);
+        float wb = (mred[wv] == -INFINITY) ? 0.f : __expf(mred[wv] - mn);
+        ss2 = ss2 * wa + sred[wv] * wb;
         mm = mn;
       }
       float l = mm + __logf(ss2);
