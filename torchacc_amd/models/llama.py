@@ -64,6 +64,9 @@ class LlamaAttention(nn.Module):
         super().__init__()
         self.cfg = cfg
         h, hk = cfg.num_attention_heads, cfg.num_key_value_heads
+        # instance attrs so tensor parallelism can shard heads per layer
+        self.num_heads = h
+        self.num_kv_heads = hk
         self.head_dim = cfg.hidden_size // h
         self.q_proj = nn.Linear(cfg.hidden_size, h * self.head_dim,
                                 bias=False)
@@ -76,7 +79,7 @@ class LlamaAttention(nn.Module):
 
     def forward(self, x, cos, sin):
         b, s, _ = x.shape
-        h, hk = self.cfg.num_attention_heads, self.cfg.num_key_value_heads
+        h, hk = self.num_heads, self.num_kv_heads
         q = self.q_proj(x).view(b, s, h, self.head_dim)
         k = self.k_proj(x).view(b, s, hk, self.head_dim)
         v = self.v_proj(x).view(b, s, hk, self.head_dim)
@@ -111,7 +114,7 @@ class LlamaAttention(nn.Module):
             o = context_parallel_2d(q, k, v, causal=True)
         else:
             raise ValueError(f"unknown cp mode {mode}")
-        return self.o_proj(o.reshape(b, s, h * self.head_dim))
+        return self.o_proj(o.reshape(b, s, h * self.head_dim))  # noqa
 
 
 class LlamaMLP(nn.Module):

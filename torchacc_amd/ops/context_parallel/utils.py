@@ -109,7 +109,24 @@ def all_to_all(x: torch.Tensor, scatter_dim: int, gather_dim: int,
         return x
     inputs = [c.contiguous() for c in x.chunk(ws, dim=scatter_dim)]
     outputs = [torch.empty_like(c) for c in inputs]
-    dist.all_to_all(outputs, inputs, group=group)
+    if dist.get_backend(group) == "gloo":
+        # gloo has no alltoall: emulate with batched isend/irecv
+        rank = dist.get_rank(group)
+        ranks = dist.get_process_group_ranks(group)
+        ops = []
+        for j in range(ws):
+            if j == rank:
+                outputs[j].copy_(inputs[j])
+                continue
+            ops.append(dist.P2POp(dist.isend, inputs[j], ranks[j],
+                                  group=group))
+            ops.append(dist.P2POp(dist.irecv, outputs[j], ranks[j],
+                                  group=group))
+        if ops:
+            for w in dist.batch_isend_irecv(ops):
+                w.wait()
+    else:
+        dist.all_to_all(outputs, inputs, group=group)
     return torch.cat(outputs, dim=gather_dim)
 
 
