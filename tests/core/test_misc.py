@@ -1,0 +1,111 @@
+"""Unit tests: async loader bucketing, cpu-offload context (CPU no-op path),
+amp scaler, lazy-API shims, fused-kernel patching."""
+import torch
+
+import torchacc_amd as ta
+
+
+def test_bucketing_loader_shapes():
+    """reference tests/core/test_bucketing_loader.py:26-61"""
+    from torchacc_amd.async_loader import (AsyncLoader, _get_closest_bucket,
+                                           _uniform_buckets)
+    assert _uniform_buckets(512, 4) == [128, 256, 384, 512]
+    assert _get_closest_bucket([128, 256, 512], 100) == 128
+    assert _get_closest_bucket([128, 256, 512], 300) == 512
+    assert _get_closest_bucket([128, 256, 512], 600) == 512
+
+    data = [{"input_ids": torch.ones(2, n, dtype=torch.long)}
+            for n in (100, 200, 500)]
+    loader = AsyncLoader(data, "cpu", buckets=[128, 256, 512])
+    shapes = [b["input_ids"].shape[-1] for b in loader]
+    assert shapes == [128, 256, 512]
+
+
+def test_async_loader_prefetch_order():
+    data = [{"x": torch.full((1, 4), i)} for i in range(10)]
+    loader = ta.AsyncLoader(data, "cpu")
+    vals = [int(b["x"][0, 0]) for b in loader]
+    assert vals == list(range(10))
+    assert len(loader) == 10
+
+
+def test_cpu_offload_context_noop_on_cpu():
+    from torchacc_amd.utils.cpu_offload import get_cpu_offload_context
+    ctx, commit = get_cpu_offload_context(2)
+    lin1 = torch.nn.Linear(8, 8)
+    lin2 = torch.nn.Linear(8, 8)
+    x = torch.randn(4, 8, requires_grad=True)
+    with ctx:
+        h = torch.relu(lin1(x))
+    h = commit(h)
+    with ctx:
+        y = lin2(h)
+    y = commit(y)
+    y.sum().backward()
+    assert x.grad is not None
+    assert lin1.weight.grad is not None
+
+
+def test_grad_scaler_cpu_disabled():
+    scaler = ta.amp.GradScaler()
+    lin = torch.nn.Linear(4, 4)
+    opt = torch.optim.SGD(lin.parameters(), lr=0.1)
+    x = torch.randn(2, 4)
+    loss = lin(x).sum()
+    scaler.scale(loss).backward()
+    scaler.step(opt)
+    scaler.update()
+    assert lin.weight.grad is not None
+
+
+def test_lazy_shims():
+    assert ta.is_lazy_tensor(torch.ones(1)) is False
+    ta.sync()
+    ta.mark_step()
+    t = torch.ones(3)
+    assert ta.mark_dynamic(t) is t
+    moved = ta.send_cpu_data_to_device({"a": t}, torch.device("cpu"))
+    assert torch.equal(moved["a"], t)
+
+
+def test_fetch_gradients():
+    lin = torch.nn.Linear(4, 4)
+    opt = torch.optim.SGD(lin.parameters(), lr=0.1)
+    lin(torch.randn(2, 4)).sum().backward()
+    grads = ta.fetch_gradients(opt)
+    assert len(grads) == 2
+
+
+def test_save_and_load(tmp_path):
+    obj = {"w": torch.randn(3)}
+    ta.save(obj, tmp_path / "x.pt")
+    back = torch.load(tmp_path / "x.pt", weights_only=False)
+    assert torch.equal(back["w"], obj["w"])
+
+
+def test_patch_amp_swaps_adamw():
+    from torchacc_amd.utils import patch
+    orig = torch.optim.AdamW
+    try:
+        patch.patch_amp()
+        from torchacc_amd.ops.adamw import AdamW
+        assert torch.optim.AdamW is AdamW
+    finally:
+        torch.optim.AdamW = orig
+
+
+def test_hf_kernel_patches_apply():
+    try:
+        import transformers  # noqa: F401
+    except ImportError:
+        return
+    from torchacc_amd.utils import patch
+    assert patch.apply_fused_kernel_patches() is True
+    from transformers.models.llama import modeling_llama
+    # patched RMSNorm runs our rms_norm
+    m = modeling_llama.LlamaRMSNorm(16)
+    x = torch.randn(2, 16)
+    y = m(x)
+    ref = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) +
+                          m.variance_epsilon) * m.weight
+    assert torch.allclose(y, ref, atol=1e-5)

@@ -21,6 +21,8 @@ from . import amp
 from . import dist
 from . import ops
 from . import utils
+from . import llm
+from .hf_trainer import accelerate_hf_trainer
 from .utils.logger import logger
 
 __version__ = "0.1.0"
@@ -105,4 +107,5 @@ __all__ = [
     "dist", "ops", "utils", "lazy_device", "sync", "mark_step",
     "is_lazy_tensor", "save", "send_cpu_data_to_device", "fetch_gradients",
     "mark_dynamic", "get_global_context", "logger",
+    "accelerate_hf_trainer", "llm",
 ]

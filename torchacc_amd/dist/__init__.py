@@ -112,5 +112,8 @@ def rendezvous(tag: str = "") -> None:
 # strategy wrappers (imported late to avoid cycles)
 from .dp import DataParallel  # noqa: E402
 from .fsdp import FullyShardedDataParallel  # noqa: E402
+from .spmd_fsdp import SpmdFullyShardedDataParallel  # noqa: E402
 from .distributed_parallel import DistributedParallel  # noqa: E402
 from .pp.pipeline import PipelineParallel  # noqa: E402
+from . import tp  # noqa: E402
+from . import state_dict_utils  # noqa: E402

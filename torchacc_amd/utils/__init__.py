@@ -1,2 +1,3 @@
-from . import checkpoint, logger, patch, utils  # noqa: F401
+from . import (checkpoint, consolidate_and_reshard_ckpts, cpu_offload,  # noqa: F401
+               decompose, import_utils, logger, patch, trace, utils)
 from .logger import logger as _logger  # noqa: F401
