@@ -96,7 +96,8 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
   const int kh = blockIdx.y % hk;
   const int b = blockIdx.y / hk;
   const int gqa = hq / hk;
-  const int kv0wg = blockIdx.x * KVWG;
+  // tile<->XCD decorrelation (see fa_fwd_kernel)
+  const int kv0wg = (int)((blockIdx.x + blockIdx.y) % gridDim.x) * KVWG;
   const int key_b = kv0wg + wid * 32;          // wave's key block base
   const int mykey = key_b + col;               // lane's key (K frag row)
   const int shift = sk - sq;
@@ -311,7 +312,8 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
   const int h = blockIdx.y % hq;
   const int b = blockIdx.y / hq;
   const int kh = h / (hq / hk);
-  const int q0wg = blockIdx.x * 256;
+  // tile<->XCD decorrelation (see fa_fwd_kernel)
+  const int q0wg = (int)((blockIdx.x + blockIdx.y) % gridDim.x) * 256;
   const int q0 = q0wg + wid * 32;
   const int qrow = q0 + col;
   const int shift = sk - sq;

@@ -63,7 +63,12 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
   const int h = blockIdx.y % hq;
   const int b = blockIdx.y / hq;
   const int kh = h / (hq / hk);
-  const int q0wg = blockIdx.x * 256;
+  // tile<->XCD decorrelation: the dispatcher places block i on XCD i%8 and
+  // blockIdx.x varies fastest, so with causal tile-skipping a direct
+  // qtile=blockIdx.x maps every heavy (diagonal) workgroup onto the same
+  // XCD; shifting by blockIdx.y balances the causal triangle (bijective
+  // per y-slice)
+  const int q0wg = (int)((blockIdx.x + blockIdx.y) % gridDim.x) * 256;
   const int q0 = q0wg + wid * 32;        // this wave's q block
   const int qrow = q0 + col;             // this lane's q row
   const int shift = sk - sq;             // bottom-right causal alignment
