@@ -75,9 +75,12 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
   constexpr int NA = D / 32;
   constexpr int QT = 32;            // q rows per staged tile
   constexpr int KVWG = 128;         // keys per workgroup (4 waves x 32)
-  // transposed-tile row stride in SHORTS: 40 (80 B = 16*5) makes 16
-  // consecutive d-rows hit 16 distinct 16B slots -> conflict-free b128
-  constexpr int TS = 40;
+  // transposed tiles [D][32]: 64B rows with a ((d>>3)&3)<<4 byte-XOR —
+  // staging lanes differ in d by multiples of 8, so the XOR must use bits
+  // >=3 of d to spread write banks; reads (b128, 16-lane groups over
+  // d = a*32+col) become conflict-free since (d&3, (d>>3)&3) is distinct
+  // per lane
+  constexpr int TS = 32;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* q_lds = reinterpret_cast<short*>(smem);            // [QT][D] swz
@@ -184,8 +187,12 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
               dv8;
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
-            qt_lds[(d0 + j) * TS + row] = qv[j];
-            dot_lds[(d0 + j) * TS + row] = dv8[j];
+            unsigned tb = (unsigned)(d0 + j) * (TS * 2) + row * 2;
+            tb ^= (unsigned)((((d0 + j) >> 3) & 3) << 4);
+            *reinterpret_cast<short*>(
+                reinterpret_cast<char*>(qt_lds) + tb) = qv[j];
+            *reinterpret_cast<short*>(
+                reinterpret_cast<char*>(dot_lds) + tb) = dv8[j];
           }
         }
         for (int r = tid; r < QT; r += 256) {
@@ -249,10 +256,12 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
           // transposed tiles (contiguous b128, stride-40 rows)
           const int d = a * 32 + col;
           const int qoff = 16 * tp + hi * 8;
+          unsigned tb = (unsigned)d * (TS * 2) + qoff * 2;
+          tb ^= (unsigned)(((d >> 3) & 3) << 4);
           bf16x8 dob = *reinterpret_cast<const bf16x8*>(
-              &dot_lds[d * TS + qoff]);
+              reinterpret_cast<const char*>(dot_lds) + tb);
           bf16x8 qbf = *reinterpret_cast<const bf16x8*>(
-              &qt_lds[d * TS + qoff]);
+              reinterpret_cast<const char*>(qt_lds) + tb);
           // dV += P^T dO ; dK += dS^T Q   (D[key=CROW][d=col])
           dvacc[a] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               pb, dob, dvacc[a], 0, 0, 0);
@@ -387,7 +396,7 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           unsigned tbyte = (unsigned)(d0 + j) * (KVB * 2) + row * 2;
-          tbyte ^= (unsigned)(((d0 + j) & 7) << 4);
+          tbyte ^= (unsigned)((((d0 + j) >> 3) & 7) << 4);
           *reinterpret_cast<short*>(reinterpret_cast<char*>(kt_lds) +
                                     tbyte) = kv8[j];
         }
@@ -454,7 +463,7 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
             const int d = a * 32 + col;
             unsigned byte = (unsigned)d * (KVB * 2) +
                             (kb * 32 + 16 * tp + hi * 8) * 2;
-            byte ^= (unsigned)((d & 7) << 4);
+            byte ^= (unsigned)(((d >> 3) & 7) << 4);
             bf16x8 ktb = *reinterpret_cast<const bf16x8*>(
                 reinterpret_cast<const char*>(kt_lds) + byte);
             dqacc[a] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
@@ -511,7 +520,7 @@ static void launch_fa_bwd(const torch::Tensor& dout, const torch::Tensor& q,
       (short*)dq.data_ptr(), b, sq, sk, hq, hk, scale, wl, wr, qlp, klp
 
   dim3 gkv((sk + 127) / 128, b * hk), bkv(256);
-  const int lds_kv = (2 * 32 * D + 2 * D * 40 + 128 * D) * 2 + 2 * 32 * 4;
+  const int lds_kv = (2 * 32 * D + 2 * D * 32 + 128 * D) * 2 + 2 * 32 * 4;
   dim3 gq((sq + 255) / 256, b * hq), bq(512);
   const int lds_q = 3 * 64 * D * 2;
 

@@ -104,40 +104,54 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
   const int t1 = (kv_hi_key + KVB - 1) / KVB;  // exclusive
   const int ntiles = t1 - t0;
 
-  // ---- staging helpers ---------------------------------------------------
+  // ---- staging helpers (T14 async split: issue loads early, write LDS
+  // late so HBM latency hides under the previous tile's compute) ----------
   const long kbase = ((long)b * sk * hk + kh) * D;  // + key*hk*D + d
-  auto stage = [&](int buf, int tile) {
+  constexpr int CHUNKS = KVB * D / 8;      // 16B chunks in tile
+  constexpr int PER_THR = CHUNKS / 512;    // chunks per thread
+  s16x8 kreg[PER_THR], vreg[PER_THR];
+
+  auto stage_load = [&](int tile) {
     const int kv0 = (t0 + tile) * KVB;
-    // K: 64 rows x D cols; 512 threads x (64*D/8/512) 16B chunks
-    constexpr int CHUNKS = KVB * D / 8;      // 16B chunks in tile
-    short* kdst = k_lds + buf * KVB * D;
-    short* vdst = vt_lds + buf * KVB * D;
 #pragma unroll
-    for (int c = tid; c < CHUNKS; c += 512) {
+    for (int i = 0; i < PER_THR; ++i) {
+      const int c = tid + i * 512;
       const int row = c / (D / 8);
       const int d0 = (c % (D / 8)) * 8;
       const int key = kv0 + row;
-      s16x8 kv8 = {0, 0, 0, 0, 0, 0, 0, 0};
-      s16x8 vv8 = {0, 0, 0, 0, 0, 0, 0, 0};
       if (key < sk) {
         const long src = kbase + (long)key * hk * D + d0;
-        kv8 = *reinterpret_cast<const s16x8*>(K + src);
-        vv8 = *reinterpret_cast<const s16x8*>(V + src);
+        kreg[i] = *reinterpret_cast<const s16x8*>(K + src);
+        vreg[i] = *reinterpret_cast<const s16x8*>(V + src);
+      } else {
+        kreg[i] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        vreg[i] = s16x8{0, 0, 0, 0, 0, 0, 0, 0};
       }
+    }
+  };
+
+  auto stage_write = [&](int buf) {
+    short* kdst = k_lds + buf * KVB * D;
+    short* vdst = vt_lds + buf * KVB * D;
+#pragma unroll
+    for (int i = 0; i < PER_THR; ++i) {
+      const int c = tid + i * 512;
+      const int row = c / (D / 8);
+      const int d0 = (c % (D / 8)) * 8;
       // K: row-major with (row&7)<<4 byte-XOR swizzle
       {
         unsigned byte = row * KROW_BYTES + d0 * 2;
         byte ^= (unsigned)((row & 7) << 4);
         *reinterpret_cast<s16x8*>(
-            reinterpret_cast<char*>(kdst) + byte) = kv8;
+            reinterpret_cast<char*>(kdst) + byte) = kreg[i];
       }
       // V^T: [D][64] with (d&7)<<4 swizzle, scalar scatter
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         unsigned byte = (unsigned)(d0 + j) * (KVB * 2) + row * 2;
-        byte ^= (unsigned)(((d0 + j) & 7) << 4);
+        byte ^= (unsigned)((((d0 + j) >> 3) & 7) << 4);
         *reinterpret_cast<short*>(
-            reinterpret_cast<char*>(vdst) + byte) = vv8[j];
+            reinterpret_cast<char*>(vdst) + byte) = vreg[i][j];
       }
     }
   };
@@ -149,7 +163,11 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
 #pragma unroll
   for (int a = 0; a < NA; ++a) oacc[a] = f32x16_(0.f);
 
-  if (ntiles > 0) stage(0, 0);
+  if (ntiles > 0) {
+    stage_load(0);
+    stage_write(0);
+    if (ntiles > 1) stage_load(1);  // in flight under tile 0's compute
+  }
   __syncthreads();
 
   for (int tile = 0; tile < ntiles; ++tile) {
@@ -264,7 +282,7 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
           // A-frag: V^T[d = a*32 + col][key = st*16 + hi*8 + j]
           const int d = a * 32 + col;
           unsigned byte = (unsigned)d * (KVB * 2) + (st * 16 + hi * 8) * 2;
-          byte ^= (unsigned)((d & 7) << 4);
+          byte ^= (unsigned)(((d >> 3) & 7) << 4);
           bf16x8 vf = *reinterpret_cast<const bf16x8*>(
               reinterpret_cast<const char*>(vbuf) + byte);
           bf16x8 pf = *reinterpret_cast<const bf16x8*>(&pb[st][0]);
@@ -274,7 +292,12 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
       }
     }
 
-    if (tile + 1 < ntiles) stage(buf ^ 1, tile + 1);
+    if (tile + 1 < ntiles) {
+      // write tile+1 (loaded one iteration ago) into the other buffer --
+      // safe: buf^1 was last read during tile-1, behind a barrier
+      stage_write(buf ^ 1);
+      if (tile + 2 < ntiles) stage_load(tile + 2);
+    }
     __syncthreads();
   }
 
