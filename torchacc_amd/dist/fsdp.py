@@ -88,10 +88,15 @@ class FlatParamUnit:
         for _, off_p in uniq.items():
             off, p = off_p
             flat[off:off + p.numel()].copy_(p.detach().reshape(-1).to(dtype))
-        # shard parameter (what the optimizer trains)
-        shard = flat[self.rank * self.shard_numel:(self.rank + 1) *
-                     self.shard_numel].clone()
-        self.shard = torch.nn.Parameter(shard)
+        # shard parameter (what the optimizer trains). ws==1: the shard IS
+        # the flat buffer (no copy, storage never freed, optimizer writes
+        # through) — the single-GPU path then has zero gather traffic.
+        if self.ws == 1:
+            self.shard = torch.nn.Parameter(flat)
+        else:
+            shard = flat[self.rank * self.shard_numel:(self.rank + 1) *
+                         self.shard_numel].clone()
+            self.shard = torch.nn.Parameter(shard)
         # full flat leaf used during compute; storage freed when resharded
         self.full_flat = flat.detach().requires_grad_(True)
         # raw alias sharing the storage but NOT the autograd version counter:
@@ -117,11 +122,9 @@ class FlatParamUnit:
     def _gather_into_full(self, async_stream):
         """All-gather shards into full_flat's storage."""
         elem = self.full_flat.element_size()
-        _alloc_storage(self.full_flat, self.padded_numel * elem)
         if self.ws == 1:
-            with torch.no_grad():
-                self._raw.copy_(self.shard.detach())
-            return
+            return  # full_flat aliases the shard param's storage
+        _alloc_storage(self.full_flat, self.padded_numel * elem)
         if async_stream is not None:
             async_stream.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(async_stream):
@@ -164,8 +167,9 @@ class FlatParamUnit:
             return
         for mod, attr, _off, _shape, _n in self.entries:
             setattr(mod, attr, None)
-        _free_storage(self.full_flat)
-        self.unsharded = False
+        if self.ws > 1:
+            _free_storage(self.full_flat)
+            self.unsharded = False
         self._views_valid = False
 
     # ---- gradient path --------------------------------------------------

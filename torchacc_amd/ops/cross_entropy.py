@@ -8,8 +8,9 @@ Replaces the reference's LigerCrossEntropyLoss / lce_forward integration
   softmax in a separate pass.
 - :func:`linear_cross_entropy`: chunks the [N, hidden] input over N, runs
   matmul -> CE -> matmul-backward per chunk so the full [N, vocab] logits
-  tensor is never materialized (what Liger's lce_forward buys, but with the
-  GEMMs on hipBLASLt and the CE on our kernel).
+  tensor is never materialized (what Liger's lce_forward buys, with the
+  GEMMs on hipBLASLt and the CE math on the CDNA4 kernel — on GPU the
+  logits chunk stays bf16 end to end).
 """
 from typing import Optional
 
@@ -24,6 +25,8 @@ class _CrossEntropy(torch.autograd.Function):
     def forward(ctx, logits, target, ignore_index):
         # logits [N, V], target [N]
         ext = dispatch(logits)
+        if logits.dtype != torch.bfloat16:
+            ext = None  # kernel is bf16-only; fp32 uses the composite path
         logits = logits.contiguous()
         if ext is not None:
             loss_sum, nvalid, lse = ext.cross_entropy_forward(
@@ -45,6 +48,8 @@ class _CrossEntropy(torch.autograd.Function):
     def backward(ctx, dloss):
         logits, target, lse, nvalid = ctx.saved_tensors
         ext = dispatch(logits)
+        if logits.dtype != torch.bfloat16:
+            ext = None
         scale = dloss.float() / nvalid.clamp_min(1).float()
         if ext is not None:
             dlogits = ext.cross_entropy_backward(
@@ -68,56 +73,84 @@ def cross_entropy(logits: torch.Tensor, target: torch.Tensor,
     return _CrossEntropy.apply(logits, target, ignore_index)
 
 
+def _chunk_ce_forward(ext, logits, target, ignore_index):
+    """(loss_sum fp32, nvalid int, lse fp32[rows]) for one chunk."""
+    if ext is not None:
+        return ext.cross_entropy_forward(logits, target, ignore_index)
+    lf = logits.float()
+    lse = torch.logsumexp(lf, dim=-1)
+    valid = target != ignore_index
+    nvalid = valid.sum()
+    tgt = target.masked_fill(~valid, 0)
+    picked = lf.gather(1, tgt.unsqueeze(1)).squeeze(1)
+    return ((lse - picked) * valid).sum(), nvalid, lse
+
+
+def _chunk_ce_backward(ext, logits, target, lse, scale, ignore_index):
+    """dlogits (logits dtype) for one chunk; scale is a 0-dim fp32 tensor."""
+    if ext is not None:
+        return ext.cross_entropy_backward(logits, target, lse, scale,
+                                          ignore_index)
+    lf = logits.float()
+    soft = torch.exp(lf - lse.unsqueeze(1))
+    valid = target != ignore_index
+    tgt = target.masked_fill(~valid, 0)
+    soft.scatter_add_(
+        1, tgt.unsqueeze(1),
+        -torch.ones_like(tgt, dtype=soft.dtype).unsqueeze(1))
+    soft *= valid.unsqueeze(1)
+    return (soft * scale).to(logits.dtype)
+
+
 class _LinearCrossEntropy(torch.autograd.Function):
     """y = CE(x @ W^T, target) without materializing full logits.
 
-    x [N, H], W [V, H]. Forward runs in chunks of rows; backward recomputes
-    the chunk logits (cheap GEMM) and accumulates dx and dW.
+    x [N, H], W [V, H]. Forward runs in row chunks (GEMM -> CE kernel, lse
+    kept per row); backward recomputes each chunk's logits (cheap GEMM) and
+    accumulates dx and dW.
     """
 
-    CHUNK = 4096
+    CHUNK = 8192
 
     @staticmethod
     def forward(ctx, x, weight, target, ignore_index):
         N = x.shape[0]
-        ctx.save_for_backward(x, weight, target)
-        ctx.ignore_index = ignore_index
-        valid = target != ignore_index
-        nvalid = valid.sum().clamp_min(1)
+        ext = dispatch(x)
+        if x.dtype != torch.bfloat16:
+            ext = None
         total = x.new_zeros((), dtype=torch.float32)
+        nvalid_t = torch.zeros((), dtype=torch.long, device=x.device)
+        lse_all = torch.empty(N, dtype=torch.float32, device=x.device)
         for s in range(0, N, _LinearCrossEntropy.CHUNK):
             e = min(N, s + _LinearCrossEntropy.CHUNK)
-            logits = (x[s:e] @ weight.t()).float()
-            lse = torch.logsumexp(logits, dim=-1)
-            v = valid[s:e]
-            tgt = target[s:e].masked_fill(~v, 0)
-            picked = logits.gather(1, tgt.unsqueeze(1)).squeeze(1)
-            total += ((lse - picked) * v).sum()
-        ctx.nvalid = nvalid
-        return total / nvalid.float()
+            logits = (x[s:e] @ weight.t()).contiguous()
+            loss_sum, nvalid, lse = _chunk_ce_forward(
+                ext, logits, target[s:e], ignore_index)
+            total += loss_sum.float()
+            nvalid_t += nvalid.long()
+            lse_all[s:e] = lse
+        ctx.save_for_backward(x, weight, target, lse_all, nvalid_t)
+        ctx.ignore_index = ignore_index
+        return total / nvalid_t.clamp_min(1).float()
 
     @staticmethod
     def backward(ctx, dloss):
-        x, weight, target = ctx.saved_tensors
+        x, weight, target, lse_all, nvalid_t = ctx.saved_tensors
         ignore_index = ctx.ignore_index
+        ext = dispatch(x)
+        if x.dtype != torch.bfloat16:
+            ext = None
         N = x.shape[0]
-        scale = (dloss / ctx.nvalid.float())
+        scale = (dloss.float() / nvalid_t.clamp_min(1).float())
         dx = torch.empty_like(x)
         dw = torch.zeros_like(weight, dtype=torch.float32)
-        valid = target != ignore_index
         for s in range(0, N, _LinearCrossEntropy.CHUNK):
             e = min(N, s + _LinearCrossEntropy.CHUNK)
-            logits = (x[s:e] @ weight.t()).float()
-            soft = torch.softmax(logits, dim=-1)
-            v = valid[s:e]
-            tgt = target[s:e].masked_fill(~v, 0)
-            soft.scatter_add_(
-                1, tgt.unsqueeze(1),
-                -torch.ones_like(tgt, dtype=soft.dtype).unsqueeze(1))
-            soft *= v.unsqueeze(1).to(soft.dtype)
-            dl = (soft * scale).to(x.dtype)
+            logits = (x[s:e] @ weight.t()).contiguous()
+            dl = _chunk_ce_backward(ext, logits, target[s:e], lse_all[s:e],
+                                    scale, ignore_index)
             dx[s:e] = dl @ weight
-            dw += (dl.t().float() @ x[s:e].float())
+            dw += (dl.t() @ x[s:e]).float()
         return dx, dw.to(weight.dtype), None, None
 
 
