@@ -31,7 +31,7 @@ def parse_args():
     p.add_argument("--model", default="llama-2-7b",
                    choices=["llama-2-7b", "llama-2-70b", "tiny"])
     p.add_argument("--seq-len", type=int, default=4096)
-    p.add_argument("--batch-size", type=int, default=2,
+    p.add_argument("--batch-size", type=int, default=8,
                    help="micro batch per GPU")
     p.add_argument("--mode", default="fsdp",
                    choices=["fsdp", "ulysses", "ring", "2d", "fsdp_tp"])

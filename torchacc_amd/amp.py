@@ -2,8 +2,10 @@
 
 Subclasses torch.amp.GradScaler; ``_unscale_grads_`` additionally all-reduces
 the found_inf flags over the PP group so every pipeline stage skips the
-optimizer step together. The flag stays device-side (no host sync) and can be
-fed to ops.AdamW's syncfree ``found_inf`` gate.
+optimizer step together. With the framework's fused AdamW the step is fully
+SYNCFREE (reference torch_xla.amp.syncfree semantics): the found_inf flag is
+handed to the optimizer kernel device-side — the update becomes a device
+no-op on overflow and the host never synchronizes.
 """
 import torch
 import torch.distributed as dist
@@ -28,3 +30,19 @@ class GradScaler(torch.amp.GradScaler):
                 for t in v:
                     dist.all_reduce(t, group=self._pp_group)
         return out
+
+    def step(self, optimizer, *args, **kwargs):
+        from .ops.adamw import AdamW as FusedAdamW
+        if not self._enabled or not isinstance(optimizer, FusedAdamW):
+            return super().step(optimizer, *args, **kwargs)
+        # syncfree path: unscale if needed, then gate in-kernel
+        state = self._per_optimizer_states[id(optimizer)]
+        from torch.amp.grad_scaler import OptState
+        if state["stage"] is OptState.READY:
+            self.unscale_(optimizer)
+        found = None
+        for t in state["found_inf_per_device"].values():
+            found = t if found is None else found + t
+        retval = optimizer.step(*args, found_inf=found, **kwargs)
+        state["stage"] = OptState.STEPPED
+        return retval

@@ -1,7 +1,9 @@
 """Ops facade (reference torchacc/ops/__init__.py:1-6)."""
 from .flash_attn import (  # noqa: F401
-    flash_attn_func, flash_attn_varlen_func, flash_attn_varlen_position_ids_xla,
-    flash_attn_varlen_xla, flash_attn_xla, spmd_flash_attn_varlen_xla)
+    flash_attn_func, flash_attn_kvpacked_xla, flash_attn_qkvpacked_xla,
+    flash_attn_varlen_func, flash_attn_varlen_position_ids_xla,
+    flash_attn_varlen_qkvpacked_xla, flash_attn_varlen_xla, flash_attn_xla,
+    spmd_flash_attn_varlen_xla)
 from .rmsnorm import RMSNorm, rms_norm  # noqa: F401
 from .rope import apply_rotary_pos_emb, build_rope_cache  # noqa: F401
 from .swiglu import swiglu  # noqa: F401

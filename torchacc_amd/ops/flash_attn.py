@@ -408,3 +408,30 @@ def flash_attn_varlen_func(q, k, v, cu_seqlens_q, cu_seqlens_k, max_seqlen_q,
 # SPMD-named alias kept for API compatibility (reference ops/flash_attn.py:66)
 def spmd_flash_attn_varlen_xla(*args, **kwargs):
     return flash_attn_varlen_xla(*args, **kwargs)
+
+
+def flash_attn_qkvpacked_xla(qkv, dropout_p=0.0, softmax_scale=None,
+                             causal=False, window_size=(-1, -1),
+                             alibi_slopes=None, deterministic=False,
+                             return_attn_probs=False):
+    """qkv [b, s, 3, h, d] (reference FlashAttnVarlenQKVPackedXla:11)."""
+    q, k, v = qkv.unbind(2)
+    return flash_attn_xla(q, k, v, dropout_p, softmax_scale, causal,
+                          window_size, alibi_slopes, deterministic,
+                          return_attn_probs)
+
+
+def flash_attn_varlen_qkvpacked_xla(qkv, attention_mask=None, **kw):
+    q, k, v = qkv.unbind(2)
+    return flash_attn_varlen_xla(q, k, v, attention_mask=attention_mask,
+                                 **kw)
+
+
+def flash_attn_kvpacked_xla(q, kv, dropout_p=0.0, softmax_scale=None,
+                            causal=False, window_size=(-1, -1),
+                            alibi_slopes=None, deterministic=False,
+                            return_attn_probs=False):
+    k, v = kv.unbind(2)
+    return flash_attn_xla(q, k, v, dropout_p, softmax_scale, causal,
+                          window_size, alibi_slopes, deterministic,
+                          return_attn_probs)
