@@ -133,8 +133,12 @@ class FlatParamUnit:
                 ev = torch.cuda.Event()
                 ev.record(async_stream)
                 self.ag_event = ev
-                # shard must not be overwritten while gather in flight
+                # shard must not be overwritten while gather in flight,
+                # and the full buffer (freed by reshard from the host
+                # thread) must not be reused before the gather stream is
+                # done with it
                 self.shard.record_stream(async_stream)
+                self._raw.record_stream(async_stream)
         else:
             dist.all_gather_into_tensor(
                 self._raw, self.shard.detach(), group=self.group)
