@@ -136,7 +136,7 @@ def test_fa_forward(causal, shape):
     ref_o, ref_lse = _ref_attention(q, k, v, scale, causal, (-1, -1))
     qg, kg, vg = _to_gpu(q, k, v)
     o, lse = ext.fa_forward(qg, kg, vg, scale, causal, -1, -1,
-                            torch.empty(0), torch.empty(0))
+                            torch.empty(0), torch.empty(0), torch.empty(0))
     do = (o.float().cpu() - ref_o.float()).abs()
     assert do.max() < 2.5e-2, f"out err {do.max()}"
     finite = torch.isfinite(ref_lse)
@@ -168,9 +168,10 @@ def test_fa_backward(causal, shape):
                                      causal, (-1, -1), None, None)
     qg, kg, vg, dog = _to_gpu(q, k, v, dout)
     o, lse = ext.fa_forward(qg, kg, vg, scale, causal, -1, -1,
-                            torch.empty(0), torch.empty(0))
+                            torch.empty(0), torch.empty(0), torch.empty(0))
     dq, dk, dv = ext.fa_backward(dog, qg, kg, vg, o, lse, scale, causal, -1,
-                                 -1, torch.empty(0), torch.empty(0))
+                                 -1, torch.empty(0), torch.empty(0),
+                                 torch.empty(0))
     for name, got, want in (("dq", dq, rdq), ("dk", dk, rdk),
                             ("dv", dv, rdv)):
         err = (got.float().cpu() - want.float()).abs().max()
@@ -192,7 +193,7 @@ def test_fa_varlen_klens():
     ref_o, _ = _ref_attention(q, k, v, scale, True, (-1, -1), lens, lens)
     qg, kg, vg = _to_gpu(q, k, v)
     o, lse = ext.fa_forward(qg, kg, vg, scale, True, -1, -1,
-                            lens.cuda(), lens.cuda())
+                            lens.cuda(), lens.cuda(), torch.empty(0))
     err = (o.float().cpu() - ref_o.float()).abs().max()
     assert err < 2.5e-2, f"varlen out err {err}"
 
@@ -210,7 +211,7 @@ def test_fa_sliding_window():
     ref_o, _ = _ref_attention(q, k, v, scale, True, (128, 0))
     qg, kg, vg = _to_gpu(q, k, v)
     o, _ = ext.fa_forward(qg, kg, vg, scale, True, 128, 0, torch.empty(0),
-                          torch.empty(0))
+                          torch.empty(0), torch.empty(0))
     err = (o.float().cpu() - ref_o.float()).abs().max()
     assert err < 2.5e-2, f"window out err {err}"
 
@@ -291,3 +292,31 @@ def test_llama70b_slice_trains():
         if l0 is None:
             l0 = float(loss)
     assert float(loss) < l0
+
+
+def test_fa_alibi():
+    from torchacc_amd.ops.flash_attn import flash_attn_xla
+    torch.manual_seed(0)
+    b, s, h, d = 2, 256, 4, 128
+    q = torch.randn(b, s, h, d)
+    k = torch.randn(b, s, h, d)
+    v = torch.randn(b, s, h, d)
+    dout = torch.randn(b, s, h, d)
+    slopes = torch.tensor([2. ** -(i + 1) for i in range(h)])
+    # CPU fp32 reference
+    qr, kr, vr = [t.clone().requires_grad_(True) for t in (q, k, v)]
+    ro = flash_attn_xla(qr, kr, vr, causal=True, alibi_slopes=slopes)
+    ro.backward(dout)
+    # GPU kernel
+    qg, kg, vg = _to_gpu(q, k, v)
+    for t in (qg, kg, vg):
+        t.requires_grad_(True)
+    og = flash_attn_xla(qg, kg, vg, causal=True,
+                        alibi_slopes=slopes.cuda())
+    og.backward(dout.to("cuda", torch.bfloat16))
+    assert (og.float().cpu() - ro.float()).abs().max() < 2.5e-2
+    for g, r in ((qg.grad, qr.grad), (kg.grad, kr.grad),
+                 (vg.grad, vr.grad)):
+        err = (g.float().cpu() - r.float()).abs().max()
+        base = r.abs().max().clamp_min(1.0)
+        assert err / base < 4e-2, err

@@ -70,7 +70,8 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
                        short* __restrict__ dK, short* __restrict__ dV,
                        int b_, int sq, int sk, int hq, int hk, float scale,
                        int wl, int wr, const int* __restrict__ q_lens,
-                       const int* __restrict__ k_lens) {
+                       const int* __restrict__ k_lens,
+                       const float* __restrict__ alibi) {
   constexpr int NT = D / 16;
   constexpr int NA = D / 32;
   constexpr int QT = 32;            // q rows per staged tile
@@ -163,6 +164,7 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
 
   for (int gh = 0; gh < gqa; ++gh) {
     const int h = kh * gqa + gh;
+    const float slope = (alibi != nullptr) ? alibi[h] : 0.f;
     for (int qt = qt0; qt < qt1; ++qt) {
       const int q0 = qt * QT;
       // ---- cooperative staging: natural (swizzled) + transposed tiles ----
@@ -303,7 +305,8 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
                       short* __restrict__ dQ, int b_, int sq, int sk, int hq,
                       int hk, float scale, int wl, int wr,
                       const int* __restrict__ q_lens,
-                      const int* __restrict__ k_lens) {
+                      const int* __restrict__ k_lens,
+                      const float* __restrict__ alibi) {
   constexpr int NT = D / 16;
   constexpr int NA = D / 32;
   constexpr int KVB = 64;
@@ -355,6 +358,7 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
     }
   }
   const bool row_ok = (qrow < qlimit);
+  const float slope = (alibi != nullptr) ? alibi[h] : 0.f;
   const float lse_q =
       row_ok ? LSE[((long)b * hq + h) * sq + qrow] : INFINITY;
   const float del_q =
@@ -446,7 +450,10 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
           if (HAS_WINDOW && wl >= 0) valid &= (key >= qrow + shift - wl);
           if (HAS_WINDOW && wr >= 0 && !CAUSAL)
             valid &= (key <= qrow + shift + wr);
-          const float p = valid ? __expf(s[kb][r] * scale - lse_q) : 0.f;
+          float sv = s[kb][r] * scale;
+          if (alibi != nullptr)
+            sv -= slope * fabsf((float)(qrow + shift - key));
+          const float p = valid ? __expf(sv - lse_q) : 0.f;
           dp[kb][r] = p * (dp[kb][r] - del_q) * scale;
         }
       }
@@ -501,7 +508,8 @@ static void launch_fa_bwd(const torch::Tensor& dout, const torch::Tensor& q,
                           const torch::Tensor& delta, torch::Tensor& dq,
                           torch::Tensor& dk, torch::Tensor& dv, float scale,
                           bool causal, int wl, int wr, const int* qlp,
-                          const int* klp, hipStream_t stream) {
+                          const int* klp, const float* alp,
+                          hipStream_t stream) {
   const int b = q.size(0), sq = q.size(1), hq = q.size(2);
   const int sk = k.size(1), hk = k.size(2);
   const bool has_window = (wl >= 0 || wr >= 0);
@@ -512,12 +520,12 @@ static void launch_fa_bwd(const torch::Tensor& dout, const torch::Tensor& q,
       (const short*)k.data_ptr(), (const short*)v.data_ptr(),                \
       lse.data_ptr<float>(), delta.data_ptr<float>(),                        \
       (short*)dk.data_ptr(), (short*)dv.data_ptr(), b, sq, sk, hq, hk,       \
-      scale, wl, wr, qlp, klp
+      scale, wl, wr, qlp, klp, alp
 #define ARGS_DQ                                                              \
   (const short*)dout.data_ptr(), (const short*)q.data_ptr(),                 \
       (const short*)k.data_ptr(), (const short*)v.data_ptr(),                \
       lse.data_ptr<float>(), delta.data_ptr<float>(),                        \
-      (short*)dq.data_ptr(), b, sq, sk, hq, hk, scale, wl, wr, qlp, klp
+      (short*)dq.data_ptr(), b, sq, sk, hq, hk, scale, wl, wr, qlp, klp, alp
 
   dim3 gkv((sk + 127) / 128, b * hk), bkv(256);
   const int lds_kv = (2 * 32 * D + 2 * D * 32 + 128 * D) * 2 + 2 * 32 * 4;
@@ -549,7 +557,8 @@ std::vector<torch::Tensor> fa_backward(torch::Tensor dout, torch::Tensor q,
                                        torch::Tensor out, torch::Tensor lse,
                                        double softmax_scale, bool causal,
                                        long wl, long wr, torch::Tensor q_lens,
-                                       torch::Tensor k_lens) {
+                                       torch::Tensor k_lens,
+                                       torch::Tensor alibi_slopes) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16,
               "fa_backward: bf16 only");
   TORCH_CHECK(dout.is_contiguous() && q.is_contiguous() &&
@@ -574,14 +583,18 @@ std::vector<torch::Tensor> fa_backward(torch::Tensor dout, torch::Tensor q,
   auto kl = k_lens.numel() ? k_lens.to(q.device(), torch::kInt32) : k_lens;
   const int* qlp = ql.numel() ? ql.data_ptr<int>() : nullptr;
   const int* klp = kl.numel() ? kl.data_ptr<int>() : nullptr;
+  auto al = alibi_slopes.numel()
+                ? alibi_slopes.to(q.device(), torch::kFloat32).contiguous()
+                : alibi_slopes;
+  const float* alp = al.numel() ? al.data_ptr<float>() : nullptr;
   if (D == 128) {
     launch_fa_bwd<128>(dout, q, k, v, lse_c, delta, dq, dk, dv,
                        (float)softmax_scale, causal, (int)wl, (int)wr, qlp,
-                       klp, stream);
+                       klp, alp, stream);
   } else {
     launch_fa_bwd<64>(dout, q, k, v, lse_c, delta, dq, dk, dv,
                       (float)softmax_scale, causal, (int)wl, (int)wr, qlp,
-                      klp, stream);
+                      klp, alp, stream);
   }
   HIP_CHECK_LAST();
   return {dq, dk, dv};

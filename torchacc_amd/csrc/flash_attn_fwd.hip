@@ -42,7 +42,8 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
                    float* __restrict__ LSE, int b_, int sq, int sk, int hq,
                    int hk, float scale, int wl, int wr,
                    const int* __restrict__ q_lens,
-                   const int* __restrict__ k_lens) {
+                   const int* __restrict__ k_lens,
+                   const float* __restrict__ alibi) {
   constexpr int NT = D / 16;   // QK^T k-steps (d slices of 16)
   constexpr int NA = D / 32;   // PV output accs (d blocks of 32)
   constexpr int KVB = 64;      // keys per tile
@@ -78,6 +79,8 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
     if (k_lens != nullptr) klimit = min(klimit, k_lens[b]);
     if (q_lens != nullptr) qlimit = min(qlimit, q_lens[b]);
   }
+  const float slope = (alibi != nullptr) ? alibi[h] : 0.f;
+  const bool has_alibi = alibi != nullptr;
 
   // ---- load Q fragments (pre-scaled) -------------------------------------
   // B-frag for swapped QK^T: lane holds Q[qrow][t*16 + hi*8 + j], j=0..7
@@ -207,7 +210,7 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
       // "full" = no per-element masking needed for ANY q row of this wave:
       // causal bound uses the wave's SMALLEST q (q0), window-left bound its
       // LARGEST q (q0+31), window-right its smallest.
-      const bool tile_full =
+      const bool tile_full = !has_alibi &&
           (!CAUSAL || kv0 + KVB - 1 <= q0 + shift) &&
           (kv0 + KVB <= klimit) &&
           (!HAS_WINDOW || wl < 0 || kv0 >= q0 + 31 + shift - wl) &&
@@ -221,6 +224,8 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
           float s = p[kb][r] * scale;
           if (!tile_full) {
             const int key = kv0 + kb * 32 + CROW(r, hi);
+            if (has_alibi)
+              s -= slope * fabsf((float)(qrow + shift - key));
             bool valid = key < klimit;
             if (CAUSAL) valid &= (key <= qrow + shift);
             if (HAS_WINDOW && wl >= 0) valid &= (key >= qrow + shift - wl);
@@ -342,7 +347,8 @@ static void launch_fa_fwd(const torch::Tensor& q, const torch::Tensor& k,
                           const torch::Tensor& v, torch::Tensor& o,
                           torch::Tensor& lse, float scale, bool causal,
                           int wl, int wr, const torch::Tensor& q_lens,
-                          const torch::Tensor& k_lens, hipStream_t stream) {
+                          const torch::Tensor& k_lens,
+                          const torch::Tensor& alibi, hipStream_t stream) {
   const int b = q.size(0), sq = q.size(1), hq = q.size(2);
   const int sk = k.size(1), hk = k.size(2);
   const bool has_window = (wl >= 0 || wr >= 0);
@@ -352,6 +358,7 @@ static void launch_fa_fwd(const torch::Tensor& q, const torch::Tensor& k,
   const int lds = 4 * 64 * D * 2;  // K + V^T, double buffered
   const int* qlp = q_lens.numel() ? q_lens.data_ptr<int>() : nullptr;
   const int* klp = k_lens.numel() ? k_lens.data_ptr<int>() : nullptr;
+  const float* alp = alibi.numel() ? alibi.data_ptr<float>() : nullptr;
 
 #define LAUNCH(CAUSAL, WIN, LENS)                                            \
   hipLaunchKernelGGL((fa_fwd_kernel<D, CAUSAL, WIN, LENS>), grid, block,     \
@@ -359,7 +366,7 @@ static void launch_fa_fwd(const torch::Tensor& q, const torch::Tensor& k,
                      (const short*)k.data_ptr(),                             \
                      (const short*)v.data_ptr(), (short*)o.data_ptr(),       \
                      lse.data_ptr<float>(), b, sq, sk, hq, hk, scale, wl,    \
-                     wr, qlp, klp)
+                     wr, qlp, klp, alp)
   if (causal) {
     if (has_window) { if (has_lens) LAUNCH(true, true, true); else LAUNCH(true, true, false); }
     else { if (has_lens) LAUNCH(true, false, true); else LAUNCH(true, false, false); }
@@ -374,7 +381,8 @@ std::vector<torch::Tensor> fa_forward(torch::Tensor q, torch::Tensor k,
                                       torch::Tensor v, double softmax_scale,
                                       bool causal, long wl, long wr,
                                       torch::Tensor q_lens,
-                                      torch::Tensor k_lens) {
+                                      torch::Tensor k_lens,
+                                      torch::Tensor alibi_slopes) {
   TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
               v.is_contiguous());
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16,
@@ -389,13 +397,18 @@ std::vector<torch::Tensor> fa_forward(torch::Tensor q, torch::Tensor k,
                            : q_lens;
   auto kl = k_lens.numel() ? k_lens.to(q.device(), torch::kInt32)
                            : k_lens;
+  auto al = alibi_slopes.numel()
+                ? alibi_slopes.to(q.device(), torch::kFloat32).contiguous()
+                : alibi_slopes;
+  TORCH_CHECK(al.numel() == 0 || al.numel() == q.size(2),
+              "alibi_slopes must be [num_heads]");
   auto stream = at::hip::getCurrentHIPStream();
   if (D == 128) {
     launch_fa_fwd<128>(q, k, v, o, lse, (float)softmax_scale, causal,
-                       (int)wl, (int)wr, ql, kl, stream);
+                       (int)wl, (int)wr, ql, kl, al, stream);
   } else {
     launch_fa_fwd<64>(q, k, v, o, lse, (float)softmax_scale, causal, (int)wl,
-                      (int)wr, ql, kl, stream);
+                      (int)wr, ql, kl, al, stream);
   }
   HIP_CHECK_LAST();
   return {o, lse};

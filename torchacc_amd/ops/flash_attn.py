@@ -39,7 +39,7 @@ def _check_qkv(q, k, v):
 
 
 def _ref_attention(q, k, v, softmax_scale, causal, window, q_lens=None,
-                   k_lens=None):
+                   k_lens=None, alibi_slopes=None):
     """fp32 composite reference; returns (out, lse[b,h,sq])."""
     b, sq, h, d = q.shape
     sk = k.shape[1]
@@ -55,6 +55,9 @@ def _ref_attention(q, k, v, softmax_scale, causal, window, q_lens=None,
     neg = torch.finfo(torch.float32).min
     iq = torch.arange(sq, device=q.device).view(1, 1, sq, 1)
     ik = torch.arange(sk, device=q.device).view(1, 1, 1, sk)
+    if alibi_slopes is not None:
+        sl = alibi_slopes.float().view(1, h, 1, 1)
+        scores = scores - sl * (ik - (iq + (sk - sq))).abs().float()
     # causal alignment: bottom-right (FA2 semantics)
     if causal:
         shift = sk - sq
@@ -89,25 +92,25 @@ class FlashAttnFunc(torch.autograd.Function):
         _check_qkv(q, k, v)
         assert dropout_p == 0.0, \
             "dropout is not yet supported by the CDNA4 kernels"
-        assert alibi_slopes is None, \
-            "alibi is not yet supported by the CDNA4 kernels"
         if softmax_scale is None:
             softmax_scale = 1.0 / math.sqrt(q.shape[-1])
         q, k, v = [t.contiguous() for t in (q, k, v)]
         ext = dispatch(q)
         wl, wr = window_size
+        al = alibi_slopes if alibi_slopes is not None else torch.empty(0)
         if ext is not None:
             out, lse = ext.fa_forward(
                 q, k, v, softmax_scale, causal, wl, wr,
                 q_lens if q_lens is not None else torch.empty(0),
-                k_lens if k_lens is not None else torch.empty(0))
+                k_lens if k_lens is not None else torch.empty(0), al)
         else:
             out, lse = _ref_attention(q, k, v, softmax_scale, causal,
-                                      (wl, wr), q_lens, k_lens)
+                                      (wl, wr), q_lens, k_lens,
+                                      alibi_slopes)
         ctx.save_for_backward(
             q, k, v, out, lse,
             q_lens if q_lens is not None else torch.empty(0),
-            k_lens if k_lens is not None else torch.empty(0))
+            k_lens if k_lens is not None else torch.empty(0), al)
         ctx.softmax_scale = softmax_scale
         ctx.causal = causal
         ctx.window = (wl, wr)
@@ -116,7 +119,7 @@ class FlashAttnFunc(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dout, _dlse):
-        q, k, v, out, lse, q_lens, k_lens = ctx.saved_tensors
+        q, k, v, out, lse, q_lens, k_lens, al = ctx.saved_tensors
         q_lens = q_lens if q_lens.numel() else None
         k_lens = k_lens if k_lens.numel() else None
         ext = dispatch(q)
@@ -127,16 +130,17 @@ class FlashAttnFunc(torch.autograd.Function):
                 dout, q, k, v, out, lse, ctx.softmax_scale, ctx.causal,
                 wl, wr,
                 q_lens if q_lens is not None else torch.empty(0),
-                k_lens if k_lens is not None else torch.empty(0))
+                k_lens if k_lens is not None else torch.empty(0), al)
         else:
             dq, dk, dv = _ref_fa_backward(dout, q, k, v, out, lse,
                                           ctx.softmax_scale, ctx.causal,
-                                          (wl, wr), q_lens, k_lens)
+                                          (wl, wr), q_lens, k_lens,
+                                          al if al.numel() else None)
         return (dq, dk, dv) + (None,) * 8
 
 
 def _ref_fa_backward(dout, q, k, v, out, lse, softmax_scale, causal, window,
-                     q_lens, k_lens):
+                     q_lens, k_lens, alibi_slopes=None):
     """Recompute-based fp32 reference backward.
 
     ``out``/``lse`` are the GLOBAL attention output / logsumexp: delta =
@@ -158,6 +162,9 @@ def _ref_fa_backward(dout, q, k, v, out, lse, softmax_scale, causal, window,
     iq = torch.arange(sq, device=q.device).view(1, 1, sq, 1)
     ik = torch.arange(sk, device=q.device).view(1, 1, 1, sk)
     shift = sk - sq
+    if alibi_slopes is not None:
+        sl = alibi_slopes.float().view(1, h, 1, 1)
+        scores = scores - sl * (ik - (iq + shift)).abs().float()
     if causal:
         scores = scores.masked_fill(ik > iq + shift, neg)
     wl, wr = window
@@ -322,7 +329,7 @@ def _varlen_fwd_gpu(ext, q, k, v, cu_q, cu_k, softmax_scale, causal, wl,
             q[qs:qe].unsqueeze(0).contiguous(),
             k[ks:ke].unsqueeze(0).contiguous(),
             v[ks:ke].unsqueeze(0).contiguous(), softmax_scale, causal, wl,
-            wr, empty, empty)
+            wr, empty, empty, empty)
         out[qs:qe] = o_i.squeeze(0)
         lse[:, qs:qe] = lse_i.squeeze(0)
     return out, lse
@@ -346,7 +353,7 @@ def _varlen_bwd_gpu(ext, dout, q, k, v, out, lse, cu_q, cu_k, softmax_scale,
             v[ks:ke].unsqueeze(0).contiguous(),
             out[qs:qe].unsqueeze(0).contiguous(),
             lse[:, qs:qe].unsqueeze(0).contiguous(), softmax_scale, causal,
-            wl, wr, empty, empty)
+            wl, wr, empty, empty, empty)
         dq[qs:qe] = dq_i.squeeze(0)
         dk[ks:ke] = dk_i.squeeze(0)
         dv[ks:ke] = dv_i.squeeze(0)
