@@ -33,14 +33,16 @@ std::vector<torch::Tensor> fa_forward(torch::Tensor q, torch::Tensor k,
                                       torch::Tensor v, double softmax_scale,
                                       bool causal, long wl, long wr,
                                       torch::Tensor q_lens,
-                                      torch::Tensor k_lens);
+                                      torch::Tensor k_lens,
+                                      torch::Tensor alibi_slopes);
 // flash_attn_bwd.hip
 std::vector<torch::Tensor> fa_backward(torch::Tensor dout, torch::Tensor q,
                                        torch::Tensor k, torch::Tensor v,
                                        torch::Tensor out, torch::Tensor lse,
                                        double softmax_scale, bool causal,
                                        long wl, long wr, torch::Tensor q_lens,
-                                       torch::Tensor k_lens);
+                                       torch::Tensor k_lens,
+                                       torch::Tensor alibi_slopes);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "torchacc_amd CDNA4 (gfx950) kernels";

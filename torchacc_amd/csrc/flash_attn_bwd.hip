@@ -239,7 +239,10 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
         if (HAS_WINDOW && wl >= 0) valid &= (mykey >= qrow + shift - wl);
         if (HAS_WINDOW && wr >= 0 && !CAUSAL)
           valid &= (mykey <= qrow + shift + wr);
-        const float p = valid ? __expf(s[r] * scale - lse_q) : 0.f;
+        float sv = s[r] * scale;
+        if (alibi != nullptr)
+          sv -= slope * fabsf((float)(qrow + shift - mykey));
+        const float p = valid ? __expf(sv - lse_q) : 0.f;
         s[r] = p;
         dp[r] = p * (dp[r] - del_q) * scale;
       }
