@@ -136,7 +136,7 @@ def test_fa_forward(causal, shape):
     ref_o, ref_lse = _ref_attention(q, k, v, scale, causal, (-1, -1))
     qg, kg, vg = _to_gpu(q, k, v)
     o, lse = ext.fa_forward(qg, kg, vg, scale, causal, -1, -1,
-                            torch.empty(0), torch.empty(0), torch.empty(0))
+                            torch.empty(0), torch.empty(0), torch.empty(0), 0.0, 0)
     do = (o.float().cpu() - ref_o.float()).abs()
     assert do.max() < 2.5e-2, f"out err {do.max()}"
     finite = torch.isfinite(ref_lse)
@@ -168,10 +168,10 @@ def test_fa_backward(causal, shape):
                                      causal, (-1, -1), None, None)
     qg, kg, vg, dog = _to_gpu(q, k, v, dout)
     o, lse = ext.fa_forward(qg, kg, vg, scale, causal, -1, -1,
-                            torch.empty(0), torch.empty(0), torch.empty(0))
+                            torch.empty(0), torch.empty(0), torch.empty(0), 0.0, 0)
     dq, dk, dv = ext.fa_backward(dog, qg, kg, vg, o, lse, scale, causal, -1,
                                  -1, torch.empty(0), torch.empty(0),
-                                 torch.empty(0))
+                                 torch.empty(0), 0.0, 0)
     for name, got, want in (("dq", dq, rdq), ("dk", dk, rdk),
                             ("dv", dv, rdv)):
         err = (got.float().cpu() - want.float()).abs().max()
@@ -193,7 +193,8 @@ def test_fa_varlen_klens():
     ref_o, _ = _ref_attention(q, k, v, scale, True, (-1, -1), lens, lens)
     qg, kg, vg = _to_gpu(q, k, v)
     o, lse = ext.fa_forward(qg, kg, vg, scale, True, -1, -1,
-                            lens.cuda(), lens.cuda(), torch.empty(0))
+                            lens.cuda(), lens.cuda(), torch.empty(0),
+                            0.0, 0)
     err = (o.float().cpu() - ref_o.float()).abs().max()
     assert err < 2.5e-2, f"varlen out err {err}"
 
@@ -211,7 +212,7 @@ def test_fa_sliding_window():
     ref_o, _ = _ref_attention(q, k, v, scale, True, (128, 0))
     qg, kg, vg = _to_gpu(q, k, v)
     o, _ = ext.fa_forward(qg, kg, vg, scale, True, 128, 0, torch.empty(0),
-                          torch.empty(0), torch.empty(0))
+                          torch.empty(0), torch.empty(0), 0.0, 0)
     err = (o.float().cpu() - ref_o.float()).abs().max()
     assert err < 2.5e-2, f"window out err {err}"
 
@@ -320,3 +321,46 @@ def test_fa_alibi():
         err = (g.float().cpu() - r.float()).abs().max()
         base = r.abs().max().clamp_min(1.0)
         assert err / base < 4e-2, err
+
+
+def test_fa_dropout():
+    """Dropout: deterministic per seed, E[out] ~ undropped out, p=0 exact,
+    backward finite + mask-consistent (ones-V trick: out rows ~ 1/keep
+    statistics)."""
+    from torchacc_amd.ops._backend import require_extension
+    ext = require_extension()
+    torch.manual_seed(0)
+    b, s, h, d = 2, 256, 4, 128
+    q = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16)
+    e = torch.empty(0)
+    scale = d ** -0.5
+    o0, lse0 = ext.fa_forward(q, k, v, scale, False, -1, -1, e, e, e,
+                              0.0, 0)
+    oa, lsea = ext.fa_forward(q, k, v, scale, False, -1, -1, e, e, e,
+                              0.3, 1234)
+    ob, _ = ext.fa_forward(q, k, v, scale, False, -1, -1, e, e, e,
+                           0.3, 1234)
+    # deterministic per seed
+    assert torch.equal(oa, ob)
+    # lse unaffected by dropout
+    assert torch.allclose(lsea, lse0, atol=1e-5)
+    # different from undropped, but unbiased on average
+    assert not torch.allclose(oa, o0, atol=1e-2)
+    m0 = o0.float().mean()
+    ma = oa.float().mean()
+    assert (ma - m0).abs() < 0.05, (float(m0), float(ma))
+    # ones-V: each output element = sum of dropped softmax row ~ mean 1
+    vones = torch.ones_like(v)
+    o1, _ = ext.fa_forward(q, k, vones, scale, False, -1, -1, e, e, e,
+                           0.3, 99)
+    mean = o1.float().mean()
+    assert (mean - 1.0).abs() < 0.03, float(mean)
+    # backward runs and is finite; zero-grad consistency at p=0
+    do = torch.randn_like(q)
+    o, lse = ext.fa_forward(q, k, v, scale, True, -1, -1, e, e, e, 0.2, 7)
+    dq, dk, dv = ext.fa_backward(do, q, k, v, o, lse, scale, True, -1, -1,
+                                 e, e, e, 0.2, 7)
+    for t in (dq, dk, dv):
+        assert torch.isfinite(t.float()).all()

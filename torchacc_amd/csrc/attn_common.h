@@ -32,3 +32,17 @@ DEVINLINE void t12_pack_frag(const V16& p16 /*16 regs*/, int tp,
     frag[u + 2] = sw[1];
   }
 }
+
+
+// counter-based dropout RNG (splitmix64): deterministic keep-decision per
+// (seed, flat attention index) — the backward kernels regenerate the same
+// mask the forward applied (reference: FA2 philox rng_state).
+DEVINLINE bool attn_dropout_keep(unsigned long long seed,
+                                 unsigned long long idx,
+                                 unsigned threshold24) {
+  unsigned long long x = seed + idx * 0x9E3779B97F4A7C15ull;
+  x ^= x >> 30; x *= 0xBF58476D1CE4E5B9ull;
+  x ^= x >> 27; x *= 0x94D049BB133111EBull;
+  x ^= x >> 31;
+  return ((unsigned)x & 0xFFFFFFu) >= threshold24;
+}

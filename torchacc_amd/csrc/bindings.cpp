@@ -34,7 +34,8 @@ std::vector<torch::Tensor> fa_forward(torch::Tensor q, torch::Tensor k,
                                       bool causal, long wl, long wr,
                                       torch::Tensor q_lens,
                                       torch::Tensor k_lens,
-                                      torch::Tensor alibi_slopes);
+                                      torch::Tensor alibi_slopes,
+                                      double p_drop, long rng_seed);
 // flash_attn_bwd.hip
 std::vector<torch::Tensor> fa_backward(torch::Tensor dout, torch::Tensor q,
                                        torch::Tensor k, torch::Tensor v,
@@ -42,7 +43,8 @@ std::vector<torch::Tensor> fa_backward(torch::Tensor dout, torch::Tensor q,
                                        double softmax_scale, bool causal,
                                        long wl, long wr, torch::Tensor q_lens,
                                        torch::Tensor k_lens,
-                                       torch::Tensor alibi_slopes);
+                                       torch::Tensor alibi_slopes,
+                                       double p_drop, long rng_seed);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "torchacc_amd CDNA4 (gfx950) kernels";

@@ -19,21 +19,10 @@
 // per-batch varlen (q_lens/k_lens), D in {64, 128}. bf16 only.
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
-#include "common.h"
-
-#define LOG2E 1.4426950408889634f
+#include "attn_common.h"
 
 typedef float f32x16_ __attribute__((ext_vector_type(16)));
-
-DEVINLINE unsigned cvt_pk_bf16(float lo, float hi) {
-  unsigned r;
-  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
-               : "=v"(r) : "v"(lo), "v"(hi));
-  return r;
-}
-
-// crow: D-matrix row for accumulator register r on lane-half hi (32x32 mfma)
-#define CROW(r, hi) (((r) & 3) + 8 * ((r) >> 2) + 4 * (hi))
+#define cvt_pk_bf16 attn_cvt_pk_bf16
 
 template <int D, bool CAUSAL, bool HAS_WINDOW, bool HAS_LENS>
 __global__ __launch_bounds__(512, 2)
@@ -43,7 +32,8 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
                    int hk, float scale, int wl, int wr,
                    const int* __restrict__ q_lens,
                    const int* __restrict__ k_lens,
-                   const float* __restrict__ alibi) {
+                   const float* __restrict__ alibi, float p_drop,
+                   unsigned long long rng_seed) {
   constexpr int NT = D / 16;   // QK^T k-steps (d slices of 16)
   constexpr int NA = D / 32;   // PV output accs (d blocks of 32)
   constexpr int KVB = 64;      // keys per tile
@@ -81,6 +71,9 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
   }
   const float slope = (alibi != nullptr) ? alibi[h] : 0.f;
   const bool has_alibi = alibi != nullptr;
+  const bool has_drop = p_drop > 0.f;
+  const float inv_keep = 1.f / (1.f - p_drop);
+  const unsigned thr24 = (unsigned)(p_drop * 16777216.f);
 
   // ---- load Q fragments (pre-scaled) -------------------------------------
   // B-frag for swapped QK^T: lane holds Q[qrow][t*16 + hi*8 + j], j=0..7
@@ -256,6 +249,21 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
       }
       lsum += __shfl_xor(lsum, 32, 64);
       l_run = l_run * alpha + lsum;
+      if (has_drop) {
+        // dropout AFTER the softmax statistics: lse is unaffected
+#pragma unroll
+        for (int kb = 0; kb < 2; ++kb) {
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int key = kv0 + kb * 32 + CROW(r, hi);
+            const unsigned long long idx =
+                (((unsigned long long)(b * hq + h) * sq + qrow)) *
+                    (unsigned long long)sk + key;
+            p[kb][r] = attn_dropout_keep(rng_seed, idx, thr24)
+                           ? p[kb][r] * inv_keep : 0.f;
+          }
+        }
+      }
 #pragma unroll
       for (int a = 0; a < NA; ++a) {
 #pragma unroll
@@ -348,7 +356,8 @@ static void launch_fa_fwd(const torch::Tensor& q, const torch::Tensor& k,
                           torch::Tensor& lse, float scale, bool causal,
                           int wl, int wr, const torch::Tensor& q_lens,
                           const torch::Tensor& k_lens,
-                          const torch::Tensor& alibi, hipStream_t stream) {
+                          const torch::Tensor& alibi, float p_drop,
+                          unsigned long long rng_seed, hipStream_t stream) {
   const int b = q.size(0), sq = q.size(1), hq = q.size(2);
   const int sk = k.size(1), hk = k.size(2);
   const bool has_window = (wl >= 0 || wr >= 0);
@@ -366,7 +375,7 @@ static void launch_fa_fwd(const torch::Tensor& q, const torch::Tensor& k,
                      (const short*)k.data_ptr(),                             \
                      (const short*)v.data_ptr(), (short*)o.data_ptr(),       \
                      lse.data_ptr<float>(), b, sq, sk, hq, hk, scale, wl,    \
-                     wr, qlp, klp, alp)
+                     wr, qlp, klp, alp, p_drop, rng_seed)
   if (causal) {
     if (has_window) { if (has_lens) LAUNCH(true, true, true); else LAUNCH(true, true, false); }
     else { if (has_lens) LAUNCH(true, false, true); else LAUNCH(true, false, false); }
@@ -382,7 +391,8 @@ std::vector<torch::Tensor> fa_forward(torch::Tensor q, torch::Tensor k,
                                       bool causal, long wl, long wr,
                                       torch::Tensor q_lens,
                                       torch::Tensor k_lens,
-                                      torch::Tensor alibi_slopes) {
+                                      torch::Tensor alibi_slopes,
+                                      double p_drop, long rng_seed) {
   TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
               v.is_contiguous());
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16,
@@ -405,10 +415,12 @@ std::vector<torch::Tensor> fa_forward(torch::Tensor q, torch::Tensor k,
   auto stream = at::hip::getCurrentHIPStream();
   if (D == 128) {
     launch_fa_fwd<128>(q, k, v, o, lse, (float)softmax_scale, causal,
-                       (int)wl, (int)wr, ql, kl, al, stream);
+                       (int)wl, (int)wr, ql, kl, al, (float)p_drop,
+                       (unsigned long long)rng_seed, stream);
   } else {
     launch_fa_fwd<64>(q, k, v, o, lse, (float)softmax_scale, causal, (int)wl,
-                      (int)wr, ql, kl, al, stream);
+                      (int)wr, ql, kl, al, (float)p_drop,
+                      (unsigned long long)rng_seed, stream);
   }
   HIP_CHECK_LAST();
   return {o, lse};

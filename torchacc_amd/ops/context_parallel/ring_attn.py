@@ -36,7 +36,7 @@ def _block_fwd(q, k, v, softmax_scale, causal):
     if ext is not None:
         return ext.fa_forward(q, k, v, softmax_scale, causal, -1, -1,
                               torch.empty(0), torch.empty(0),
-                              torch.empty(0))
+                              torch.empty(0), 0.0, 0)
     return _ref_attention(q, k, v, softmax_scale, causal, (-1, -1))
 
 
@@ -45,7 +45,7 @@ def _block_bwd(dout, q, k, v, out, lse, softmax_scale, causal):
     if ext is not None:
         return ext.fa_backward(dout, q, k, v, out, lse, softmax_scale,
                                causal, -1, -1, torch.empty(0),
-                               torch.empty(0), torch.empty(0))
+                               torch.empty(0), torch.empty(0), 0.0, 0)
     return _ref_fa_backward(dout, q, k, v, out, lse, softmax_scale, causal,
                             (-1, -1), None, None)
 

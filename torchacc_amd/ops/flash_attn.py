@@ -90,19 +90,25 @@ class FlashAttnFunc(torch.autograd.Function):
     def forward(ctx, q, k, v, dropout_p, softmax_scale, causal, window_size,
                 alibi_slopes, deterministic, q_lens, k_lens):
         _check_qkv(q, k, v)
-        assert dropout_p == 0.0, \
-            "dropout is not yet supported by the CDNA4 kernels"
         if softmax_scale is None:
             softmax_scale = 1.0 / math.sqrt(q.shape[-1])
         q, k, v = [t.contiguous() for t in (q, k, v)]
         ext = dispatch(q)
         wl, wr = window_size
         al = alibi_slopes if alibi_slopes is not None else torch.empty(0)
+        seed = 0
+        if dropout_p > 0.0:
+            assert ext is not None, \
+                "attention dropout runs only on the CDNA4 kernels (GPU)"
+            # host-RNG seed: reproducible under torch.manual_seed; the
+            # backward kernels regenerate the same keep-mask from it
+            seed = int(torch.randint(0, 2 ** 62, (1,)).item())
         if ext is not None:
             out, lse = ext.fa_forward(
                 q, k, v, softmax_scale, causal, wl, wr,
                 q_lens if q_lens is not None else torch.empty(0),
-                k_lens if k_lens is not None else torch.empty(0), al)
+                k_lens if k_lens is not None else torch.empty(0), al,
+                dropout_p, seed)
         else:
             out, lse = _ref_attention(q, k, v, softmax_scale, causal,
                                       (wl, wr), q_lens, k_lens,
@@ -115,6 +121,7 @@ class FlashAttnFunc(torch.autograd.Function):
         ctx.causal = causal
         ctx.window = (wl, wr)
         ctx.deterministic = deterministic
+        ctx.dropout = (dropout_p, seed)
         return out, lse
 
     @staticmethod
@@ -126,11 +133,13 @@ class FlashAttnFunc(torch.autograd.Function):
         dout = dout.contiguous()
         wl, wr = ctx.window
         if ext is not None:
+            p_drop, seed = ctx.dropout
             dq, dk, dv = ext.fa_backward(
                 dout, q, k, v, out, lse, ctx.softmax_scale, ctx.causal,
                 wl, wr,
                 q_lens if q_lens is not None else torch.empty(0),
-                k_lens if k_lens is not None else torch.empty(0), al)
+                k_lens if k_lens is not None else torch.empty(0), al,
+                p_drop, seed)
         else:
             dq, dk, dv = _ref_fa_backward(dout, q, k, v, out, lse,
                                           ctx.softmax_scale, ctx.causal,
@@ -329,7 +338,7 @@ def _varlen_fwd_gpu(ext, q, k, v, cu_q, cu_k, softmax_scale, causal, wl,
             q[qs:qe].unsqueeze(0).contiguous(),
             k[ks:ke].unsqueeze(0).contiguous(),
             v[ks:ke].unsqueeze(0).contiguous(), softmax_scale, causal, wl,
-            wr, empty, empty, empty)
+            wr, empty, empty, empty, 0.0, 0)
         out[qs:qe] = o_i.squeeze(0)
         lse[:, qs:qe] = lse_i.squeeze(0)
     return out, lse
@@ -353,7 +362,7 @@ def _varlen_bwd_gpu(ext, dout, q, k, v, out, lse, cu_q, cu_k, softmax_scale,
             v[ks:ke].unsqueeze(0).contiguous(),
             out[qs:qe].unsqueeze(0).contiguous(),
             lse[:, qs:qe].unsqueeze(0).contiguous(), softmax_scale, causal,
-            wl, wr, empty, empty, empty)
+            wl, wr, empty, empty, empty, 0.0, 0)
         dq[qs:qe] = dq_i.squeeze(0)
         dk[ks:ke] = dk_i.squeeze(0)
         dv[ks:ke] = dv_i.squeeze(0)
