@@ -11,6 +11,14 @@ std::vector<torch::Tensor> rmsnorm_backward(torch::Tensor dy, torch::Tensor x,
 std::vector<torch::Tensor> rope_forward(torch::Tensor q, torch::Tensor k,
                                         torch::Tensor cos, torch::Tensor sin);
 torch::Tensor swiglu_forward(torch::Tensor g, torch::Tensor u);
+std::vector<torch::Tensor> add_rmsnorm_forward(torch::Tensor x,
+                                               torch::Tensor resid,
+                                               torch::Tensor w, double eps);
+std::vector<torch::Tensor> add_rmsnorm_backward(torch::Tensor dy,
+                                                torch::Tensor dresid,
+                                                torch::Tensor r_saved,
+                                                torch::Tensor w,
+                                                torch::Tensor inv_rms);
 std::vector<torch::Tensor> swiglu_backward(torch::Tensor dy, torch::Tensor g,
                                            torch::Tensor u);
 // cross_entropy.hip
@@ -52,6 +60,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_backward", &rmsnorm_backward);
   m.def("rope_forward", &rope_forward);
   m.def("swiglu_forward", &swiglu_forward);
+  m.def("add_rmsnorm_forward", &add_rmsnorm_forward);
+  m.def("add_rmsnorm_backward", &add_rmsnorm_backward);
   m.def("swiglu_backward", &swiglu_backward);
   m.def("cross_entropy_forward", &cross_entropy_forward);
   m.def("cross_entropy_backward", &cross_entropy_backward);

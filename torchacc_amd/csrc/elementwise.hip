@@ -103,6 +103,117 @@ __global__ void rmsnorm_bwd_kernel(const short* __restrict__ dy,
     atomicAdd(&dw[i], dw_acc[i]);
 }
 
+// fused residual add + RMSNorm: r = x + resid_in; y = rmsnorm(r) * w.
+// Saves one full read+write of the residual stream per call vs separate
+// add and norm kernels (and the separate add's backward elementwise).
+__global__ void add_rmsnorm_fwd_kernel(const short* __restrict__ x,
+                                       const short* __restrict__ resid_in,
+                                       const short* __restrict__ w,
+                                       short* __restrict__ y,
+                                       short* __restrict__ resid_out,
+                                       float* __restrict__ inv_rms, int rows,
+                                       int H, float eps, bool has_resid) {
+  __shared__ float red[16];
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const short* xr = x + (long)row * H;
+    const short* rr = resid_in + (long)row * H;
+    short* yr = y + (long)row * H;
+    short* ro = resid_out + (long)row * H;
+    float ss = 0.f;
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+      s16x8 v = *reinterpret_cast<const s16x8*>(xr + i);
+      s16x8 o;
+      if (has_resid) {
+        s16x8 rv = *reinterpret_cast<const s16x8*>(rr + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = bf16_to_f32(v[j]) + bf16_to_f32(rv[j]);
+          o[j] = f32_to_bf16(f);
+          ss += f * f;
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = bf16_to_f32(v[j]);
+          o[j] = v[j];
+          ss += f * f;
+        }
+      }
+      *reinterpret_cast<s16x8*>(ro + i) = o;
+    }
+    ss = block_reduce_sum<4>(ss, red);
+    float r = rsqrtf(ss / H + eps);
+    if (threadIdx.x == 0) inv_rms[row] = r;
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+      s16x8 v = *reinterpret_cast<const s16x8*>(ro + i);
+      s16x8 wv = *reinterpret_cast<const s16x8*>(w + i);
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = f32_to_bf16(bf16_to_f32(v[j]) * r * bf16_to_f32(wv[j]));
+      *reinterpret_cast<s16x8*>(yr + i) = o;
+    }
+  }
+}
+
+// backward: dx = rmsnorm_dx(dy) + dresid (grad into the residual stream
+// from downstream); dw accumulated as in rmsnorm_bwd.
+__global__ void add_rmsnorm_bwd_kernel(const short* __restrict__ dy,
+                                       const short* __restrict__ dresid,
+                                       const short* __restrict__ r_saved,
+                                       const short* __restrict__ w,
+                                       const float* __restrict__ inv_rms,
+                                       short* __restrict__ dx,
+                                       float* __restrict__ dw, int rows,
+                                       int H, bool has_dresid) {
+  extern __shared__ float smem[];
+  float* dw_acc = smem;
+  float* red = smem + H;
+  for (int i = threadIdx.x; i < H; i += blockDim.x) dw_acc[i] = 0.f;
+  __syncthreads();
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const short* xr = r_saved + (long)row * H;
+    const short* dyr = dy + (long)row * H;
+    const short* drr = dresid + (long)row * H;
+    short* dxr = dx + (long)row * H;
+    const float rinv = inv_rms[row];
+    float dot = 0.f;
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+      s16x8 xv = *reinterpret_cast<const s16x8*>(xr + i);
+      s16x8 dv = *reinterpret_cast<const s16x8*>(dyr + i);
+      s16x8 wv = *reinterpret_cast<const s16x8*>(w + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xh = bf16_to_f32(xv[j]) * rinv;
+        dot += bf16_to_f32(wv[j]) * bf16_to_f32(dv[j]) * xh;
+      }
+    }
+    dot = block_reduce_sum<4>(dot, red) / H;
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+      s16x8 xv = *reinterpret_cast<const s16x8*>(xr + i);
+      s16x8 dv = *reinterpret_cast<const s16x8*>(dyr + i);
+      s16x8 wv = *reinterpret_cast<const s16x8*>(w + i);
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xh = bf16_to_f32(xv[j]) * rinv;
+        float dyf = bf16_to_f32(dv[j]);
+        float g = rinv * (bf16_to_f32(wv[j]) * dyf - xh * dot);
+        if (has_dresid) {
+          s16x8 drv = *reinterpret_cast<const s16x8*>(drr + i);
+          g += bf16_to_f32(drv[j]);
+        }
+        o[j] = f32_to_bf16(g);
+        dw_acc[i + j] += dyf * xh;
+      }
+      *reinterpret_cast<s16x8*>(dxr + i) = o;
+    }
+    __syncthreads();
+  }
+  for (int i = threadIdx.x; i < H; i += blockDim.x)
+    atomicAdd(&dw[i], dw_acc[i]);
+}
+
 // fp32 variants (CPU-parity/debug path; also used when model is fp32)
 __global__ void rmsnorm_fwd_kernel_f32(const float* __restrict__ x,
                                        const float* __restrict__ w,
@@ -333,4 +444,56 @@ std::vector<torch::Tensor> swiglu_backward(torch::Tensor dy, torch::Tensor g,
                      (short*)dg.data_ptr(), (short*)du.data_ptr(), n8);
   HIP_CHECK_LAST();
   return {dg, du};
+}
+
+
+std::vector<torch::Tensor> add_rmsnorm_forward(torch::Tensor x,
+                                               torch::Tensor resid,
+                                               torch::Tensor w, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16,
+              "add_rmsnorm: bf16 only on GPU");
+  const int H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0);
+  const long rows = x.numel() / H;
+  const bool has_resid = resid.numel() > 0;
+  auto y = torch::empty_like(x);
+  auto resid_out = torch::empty_like(x);
+  auto inv_rms = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(add_rmsnorm_fwd_kernel,
+                     dim3(std::min<long>(rows, 2048)), dim3(256), 0, stream,
+                     (const short*)x.data_ptr(),
+                     has_resid ? (const short*)resid.data_ptr()
+                               : (const short*)x.data_ptr(),
+                     (const short*)w.data_ptr(), (short*)y.data_ptr(),
+                     (short*)resid_out.data_ptr(), inv_rms.data_ptr<float>(),
+                     (int)rows, H, (float)eps, has_resid);
+  HIP_CHECK_LAST();
+  return {y, resid_out, inv_rms};
+}
+
+std::vector<torch::Tensor> add_rmsnorm_backward(torch::Tensor dy,
+                                                torch::Tensor dresid,
+                                                torch::Tensor r_saved,
+                                                torch::Tensor w,
+                                                torch::Tensor inv_rms) {
+  const int H = r_saved.size(-1);
+  const long rows = r_saved.numel() / H;
+  const bool has_dresid = dresid.numel() > 0;
+  auto dx = torch::empty_like(r_saved);
+  auto dw32 = torch::zeros({H}, r_saved.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  const int lds = (H + 16) * sizeof(float);
+  hipLaunchKernelGGL(add_rmsnorm_bwd_kernel,
+                     dim3(std::min<long>(rows, 512)), dim3(256), lds, stream,
+                     (const short*)dy.data_ptr(),
+                     has_dresid ? (const short*)dresid.data_ptr()
+                                : (const short*)dy.data_ptr(),
+                     (const short*)r_saved.data_ptr(),
+                     (const short*)w.data_ptr(), inv_rms.data_ptr<float>(),
+                     (short*)dx.data_ptr(), dw32.data_ptr<float>(),
+                     (int)rows, H, has_dresid);
+  HIP_CHECK_LAST();
+  return {dx, dw32.to(w.scalar_type())};
 }

@@ -19,7 +19,7 @@ import torch.nn as nn
 
 from ..ops.cross_entropy import linear_cross_entropy
 from ..ops.flash_attn import flash_attn_xla
-from ..ops.rmsnorm import RMSNorm
+from ..ops.rmsnorm import RMSNorm, fused_add_rms_norm
 from ..ops.rope import apply_rotary_pos_emb, build_rope_cache
 from ..ops.swiglu import swiglu
 
@@ -133,6 +133,10 @@ class LlamaMLP(nn.Module):
 
 
 class LlamaDecoderLayer(nn.Module):
+    """Residual-stream form: the layer receives (residual, delta) with
+    hidden = residual + delta and returns the same pair — every residual
+    add is fused into the following RMSNorm (one kernel: add + norm +
+    residual-grad accumulation in backward)."""
 
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
@@ -142,10 +146,16 @@ class LlamaDecoderLayer(nn.Module):
                                                 cfg.rms_norm_eps)
         self.mlp = LlamaMLP(cfg)
 
-    def forward(self, x, cos, sin):
-        x = x + self.self_attn(self.input_layernorm(x), cos, sin)
-        x = x + self.mlp(self.post_attention_layernorm(x))
-        return x
+    def forward(self, residual, delta, cos, sin):
+        y1, resid = fused_add_rms_norm(
+            delta, residual, self.input_layernorm.weight,
+            self.input_layernorm.variance_epsilon)
+        a = self.self_attn(y1, cos, sin)
+        y2, resid2 = fused_add_rms_norm(
+            a, resid, self.post_attention_layernorm.weight,
+            self.post_attention_layernorm.variance_epsilon)
+        m = self.mlp(y2)
+        return resid2, m
 
 
 class LlamaForCausalLM(nn.Module):
@@ -177,11 +187,13 @@ class LlamaForCausalLM(nn.Module):
     def forward(self, input_ids: torch.Tensor,
                 labels: Optional[torch.Tensor] = None,
                 attention_mask=None):
-        x = self.embed_tokens(input_ids)
+        delta = self.embed_tokens(input_ids)
+        residual = None
         cos, sin = self.rope_cos, self.rope_sin
         for layer in self.layers:
-            x = layer(x, cos, sin)
-        x = self.norm(x)
+            residual, delta = layer(residual, delta, cos, sin)
+        x, _ = fused_add_rms_norm(delta, residual, self.norm.weight,
+                                  self.norm.variance_epsilon)
         if labels is not None:
             # shift: predict token t+1 from position t; fused linear+CE
             hs = x[:, :-1, :].reshape(-1, x.shape[-1])

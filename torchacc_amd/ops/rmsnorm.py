@@ -68,3 +68,63 @@ class RMSNorm(torch.nn.Module):
 
     def extra_repr(self):
         return f"{self.weight.shape[0]}, eps={self.variance_epsilon}"
+
+
+class _AddRMSNorm(torch.autograd.Function):
+    """Fused residual add + RMSNorm: r = x + residual; y = rmsnorm(r) * w.
+    Returns (y, r); the residual stream r flows to the next layer so its
+    incoming grad is fused into dx in backward (one kernel instead of an
+    add + a norm + an elementwise grad-add)."""
+
+    @staticmethod
+    def forward(ctx, x, residual, weight, eps):
+        ext = dispatch(x)
+        x = x.contiguous()
+        has_resid = residual is not None
+        if ext is not None:
+            y, r, inv_rms = ext.add_rmsnorm_forward(
+                x, residual.contiguous() if has_resid else
+                torch.empty(0, device=x.device, dtype=x.dtype),
+                weight, eps)
+        else:
+            r = (x + residual) if has_resid else x
+            y, inv_rms = _ref_rms_forward(r, weight, eps)
+            inv_rms = inv_rms.squeeze(-1)
+        ctx.save_for_backward(r, weight, inv_rms)
+        ctx.eps = eps
+        ctx.has_resid = has_resid
+        return y, r
+
+    @staticmethod
+    def backward(ctx, dy, dresid):
+        r, weight, inv_rms = ctx.saved_tensors
+        ext = dispatch(r)
+        dy = dy.contiguous()
+        if ext is not None:
+            dx, dw = ext.add_rmsnorm_backward(
+                dy, dresid.contiguous() if dresid is not None else
+                torch.empty(0, device=r.device, dtype=r.dtype),
+                r, weight, inv_rms)
+        else:
+            rf = r.float()
+            dyf = dy.float()
+            wf = weight.float()
+            ri = inv_rms.unsqueeze(-1)
+            xhat = rf * ri
+            wdy = dyf * wf
+            ddot = (wdy * xhat).mean(-1, keepdim=True)
+            dx = (ri * (wdy - xhat * ddot))
+            if dresid is not None:
+                dx = dx + dresid.float()
+            dx = dx.to(r.dtype)
+            dw = (dyf * xhat).reshape(-1, r.shape[-1]).sum(0).to(
+                weight.dtype)
+        dres = dx if ctx.has_resid else None
+        return dx, dres, dw, None
+
+
+def fused_add_rms_norm(x: torch.Tensor, residual, weight: torch.Tensor,
+                       eps: float = 1e-6):
+    """(y, new_residual) = (rmsnorm(x + residual) * w, x + residual);
+    residual may be None (plain norm, r = x)."""
+    return _AddRMSNorm.apply(x, residual, weight, eps)

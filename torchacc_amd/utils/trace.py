@@ -24,7 +24,7 @@ def _leaf_fns():
     for modname, names in (
             ("torchacc_amd.ops.flash_attn",
              ("flash_attn_xla", "flash_attn_varlen_xla", "flash_attn_func")),
-            ("torchacc_amd.ops.rmsnorm", ("rms_norm",)),
+            ("torchacc_amd.ops.rmsnorm", ("rms_norm", "fused_add_rms_norm")),
             ("torchacc_amd.ops.rope", ("apply_rotary_pos_emb",)),
             ("torchacc_amd.ops.swiglu", ("swiglu",)),
             ("torchacc_amd.ops.cross_entropy",
