@@ -70,9 +70,7 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
                        short* __restrict__ dK, short* __restrict__ dV,
                        int b_, int sq, int sk, int hq, int hk, float scale,
                        int wl, int wr, const int* __restrict__ q_lens,
-                       const int* __restrict__ k_lens,
-                       const float* __restrict__ alibi, float p_drop,
-                       unsigned long long rng_seed) {
+                       const int* __restrict__ k_lens) {
   constexpr int NT = D / 16;
   constexpr int NA = D / 32;
   constexpr int QT = 32;            // q rows per staged tile
@@ -114,9 +112,6 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
     if (k_lens != nullptr) klimit = min(klimit, k_lens[b]);
     if (q_lens != nullptr) qlimit = min(qlimit, q_lens[b]);
   }
-  const bool has_drop = p_drop > 0.f;
-  const float inv_keep = 1.f / (1.f - p_drop);
-  const unsigned thr24 = (unsigned)(p_drop * 16777216.f);
 
   // K fragments in registers; V staged once in LDS for the whole workgroup
   bf16x8 kfrag[NT];
@@ -168,7 +163,6 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
 
   for (int gh = 0; gh < gqa; ++gh) {
     const int h = kh * gqa + gh;
-    const float slope = (alibi != nullptr) ? alibi[h] : 0.f;
     for (int qt = qt0; qt < qt1; ++qt) {
       const int q0 = qt * QT;
       // ---- cooperative staging: natural (swizzled) + transposed tiles ----
@@ -244,23 +238,9 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
         if (HAS_WINDOW && wl >= 0) valid &= (mykey >= qrow + shift - wl);
         if (HAS_WINDOW && wr >= 0 && !CAUSAL)
           valid &= (mykey <= qrow + shift + wr);
-        float sv = s[r] * scale;
-        if (alibi != nullptr)
-          sv -= slope * fabsf((float)(qrow + shift - mykey));
-        const float p = valid ? __expf(sv - lse_q) : 0.f;
-        if (has_drop) {
-          const unsigned long long idx =
-              (((unsigned long long)(b * hq + h) * sq + qrow)) *
-                  (unsigned long long)sk + mykey;
-          const float keep =
-              attn_dropout_keep(rng_seed, idx, thr24) ? inv_keep : 0.f;
-          // dV uses the dropped P; dS = P * (masked dP - delta)
-          s[r] = p * keep;
-          dp[r] = p * (dp[r] * keep - del_q) * scale;
-        } else {
-          s[r] = p;
-          dp[r] = p * (dp[r] - del_q) * scale;
-        }
+        const float p = valid ? __expf(s[r] * scale - lse_q) : 0.f;
+        s[r] = p;
+        dp[r] = p * (dp[r] - del_q) * scale;
       }
       // A-frags over the q k-dim; step tp covers q rows 16tp..16tp+15
 #pragma unroll
@@ -323,9 +303,7 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
                       short* __restrict__ dQ, int b_, int sq, int sk, int hq,
                       int hk, float scale, int wl, int wr,
                       const int* __restrict__ q_lens,
-                      const int* __restrict__ k_lens,
-                      const float* __restrict__ alibi, float p_drop,
-                      unsigned long long rng_seed) {
+                      const int* __restrict__ k_lens) {
   constexpr int NT = D / 16;
   constexpr int NA = D / 32;
   constexpr int KVB = 64;
@@ -377,10 +355,6 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
     }
   }
   const bool row_ok = (qrow < qlimit);
-  const bool has_drop = p_drop > 0.f;
-  const float inv_keep = 1.f / (1.f - p_drop);
-  const unsigned thr24 = (unsigned)(p_drop * 16777216.f);
-  const float slope = (alibi != nullptr) ? alibi[h] : 0.f;
   const float lse_q =
       row_ok ? LSE[((long)b * hq + h) * sq + qrow] : INFINITY;
   const float del_q =
@@ -472,18 +446,8 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
           if (HAS_WINDOW && wl >= 0) valid &= (key >= qrow + shift - wl);
           if (HAS_WINDOW && wr >= 0 && !CAUSAL)
             valid &= (key <= qrow + shift + wr);
-          float sv = s[kb][r] * scale;
-          if (alibi != nullptr)
-            sv -= slope * fabsf((float)(qrow + shift - key));
-          const float p = valid ? __expf(sv - lse_q) : 0.f;
-          float dpv = dp[kb][r];
-          if (has_drop) {
-            const unsigned long long idx =
-                (((unsigned long long)(b * hq + h) * sq + qrow)) *
-                    (unsigned long long)sk + key;
-            dpv *= attn_dropout_keep(rng_seed, idx, thr24) ? inv_keep : 0.f;
-          }
-          dp[kb][r] = p * (dpv - del_q) * scale;
+          const float p = valid ? __expf(s[kb][r] * scale - lse_q) : 0.f;
+          dp[kb][r] = p * (dp[kb][r] - del_q) * scale;
         }
       }
       // dQ[q][d] += dS[q][key] K[key][d]
@@ -537,9 +501,7 @@ static void launch_fa_bwd(const torch::Tensor& dout, const torch::Tensor& q,
                           const torch::Tensor& delta, torch::Tensor& dq,
                           torch::Tensor& dk, torch::Tensor& dv, float scale,
                           bool causal, int wl, int wr, const int* qlp,
-                          const int* klp, const float* alp, float p_drop,
-                          unsigned long long rng_seed,
-                          hipStream_t stream) {
+                          const int* klp, hipStream_t stream) {
   const int b = q.size(0), sq = q.size(1), hq = q.size(2);
   const int sk = k.size(1), hk = k.size(2);
   const bool has_window = (wl >= 0 || wr >= 0);
@@ -550,13 +512,12 @@ static void launch_fa_bwd(const torch::Tensor& dout, const torch::Tensor& q,
       (const short*)k.data_ptr(), (const short*)v.data_ptr(),                \
       lse.data_ptr<float>(), delta.data_ptr<float>(),                        \
       (short*)dk.data_ptr(), (short*)dv.data_ptr(), b, sq, sk, hq, hk,       \
-      scale, wl, wr, qlp, klp, alp, p_drop, rng_seed
+      scale, wl, wr, qlp, klp
 #define ARGS_DQ                                                              \
   (const short*)dout.data_ptr(), (const short*)q.data_ptr(),                 \
       (const short*)k.data_ptr(), (const short*)v.data_ptr(),                \
       lse.data_ptr<float>(), delta.data_ptr<float>(),                        \
-      (short*)dq.data_ptr(), b, sq, sk, hq, hk, scale, wl, wr, qlp, klp,    \
-      alp, p_drop, rng_seed
+      (short*)dq.data_ptr(), b, sq, sk, hq, hk, scale, wl, wr, qlp, klp
 
   dim3 gkv((sk + 127) / 128, b * hk), bkv(256);
   const int lds_kv = (2 * 32 * D + 2 * D * 32 + 128 * D) * 2 + 2 * 32 * 4;
@@ -583,6 +544,12 @@ static void launch_fa_bwd(const torch::Tensor& dout, const torch::Tensor& q,
 #undef ARGS_DQ
 }
 
+std::vector<torch::Tensor> fa_backward_extra(
+    torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor out, torch::Tensor lse, double softmax_scale, bool causal,
+    long wl, long wr, torch::Tensor q_lens, torch::Tensor k_lens,
+    torch::Tensor alibi_slopes, double p_drop, long rng_seed);
+
 std::vector<torch::Tensor> fa_backward(torch::Tensor dout, torch::Tensor q,
                                        torch::Tensor k, torch::Tensor v,
                                        torch::Tensor out, torch::Tensor lse,
@@ -591,6 +558,11 @@ std::vector<torch::Tensor> fa_backward(torch::Tensor dout, torch::Tensor q,
                                        torch::Tensor k_lens,
                                        torch::Tensor alibi_slopes,
                                        double p_drop, long rng_seed) {
+  if (alibi_slopes.numel() > 0 || p_drop > 0.0) {
+    return fa_backward_extra(dout, q, k, v, out, lse, softmax_scale, causal,
+                             wl, wr, q_lens, k_lens, alibi_slopes, p_drop,
+                             rng_seed);
+  }
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16,
               "fa_backward: bf16 only");
   TORCH_CHECK(dout.is_contiguous() && q.is_contiguous() &&
@@ -615,20 +587,14 @@ std::vector<torch::Tensor> fa_backward(torch::Tensor dout, torch::Tensor q,
   auto kl = k_lens.numel() ? k_lens.to(q.device(), torch::kInt32) : k_lens;
   const int* qlp = ql.numel() ? ql.data_ptr<int>() : nullptr;
   const int* klp = kl.numel() ? kl.data_ptr<int>() : nullptr;
-  auto al = alibi_slopes.numel()
-                ? alibi_slopes.to(q.device(), torch::kFloat32).contiguous()
-                : alibi_slopes;
-  const float* alp = al.numel() ? al.data_ptr<float>() : nullptr;
   if (D == 128) {
     launch_fa_bwd<128>(dout, q, k, v, lse_c, delta, dq, dk, dv,
                        (float)softmax_scale, causal, (int)wl, (int)wr, qlp,
-                       klp, alp, (float)p_drop,
-                       (unsigned long long)rng_seed, stream);
+                       klp, stream);
   } else {
     launch_fa_bwd<64>(dout, q, k, v, lse_c, delta, dq, dk, dv,
                       (float)softmax_scale, causal, (int)wl, (int)wr, qlp,
-                      klp, alp, (float)p_drop,
-                      (unsigned long long)rng_seed, stream);
+                      klp, stream);
   }
   HIP_CHECK_LAST();
   return {dq, dk, dv};

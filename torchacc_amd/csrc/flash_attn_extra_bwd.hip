@@ -1,3 +1,7 @@
+// Full-featured flash-attention BACKWARD (alibi + dropout), separate
+// translation unit: the vanilla hot path (flash_attn_fwd.hip) stays at its
+// tuned register budget; these variants tolerate spills (guide rule 19:
+// co-compiled template variants perturb each other's codegen).
 // Flash-attention v2 BACKWARD for gfx950 (CDNA4 MFMA).
 //
 // Three kernels (recompute variant — deterministic, no atomics):
@@ -30,7 +34,7 @@
 // ---------------------------------------------------------------------------
 // 1. preprocess: delta = rowsum(dO * O)   [b,h,sq] fp32
 // ---------------------------------------------------------------------------
-__global__ void fa_bwd_preprocess_kernel(const short* __restrict__ dO,
+__global__ void fa_bwd_preprocess_kernel_x(const short* __restrict__ dO,
                                          const short* __restrict__ O,
                                          float* __restrict__ delta, int b,
                                          int sq, int hq, int D) {
@@ -58,9 +62,9 @@ __global__ void fa_bwd_preprocess_kernel(const short* __restrict__ dO,
 // ---------------------------------------------------------------------------
 // 2. dK/dV kernel
 // ---------------------------------------------------------------------------
-template <int D, bool CAUSAL, bool HAS_WINDOW, bool HAS_LENS>
+template <int D, bool CAUSAL, bool HAS_WINDOW, bool HAS_LENS, bool HAS_EXTRA>
 __global__ __launch_bounds__(256, 2)
-void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
+void fa_bwd_dkv_kernel_x(const short* __restrict__ dOut,
                        const short* __restrict__ Q,
                        const short* __restrict__ K,
                        const short* __restrict__ V,
@@ -69,7 +73,9 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
                        short* __restrict__ dK, short* __restrict__ dV,
                        int b_, int sq, int sk, int hq, int hk, float scale,
                        int wl, int wr, const int* __restrict__ q_lens,
-                       const int* __restrict__ k_lens) {
+                       const int* __restrict__ k_lens,
+                       const float* __restrict__ alibi, float p_drop,
+                       unsigned long long rng_seed) {
   constexpr int NT = D / 16;
   constexpr int NA = D / 32;
   constexpr int QT = 32;            // q rows per staged tile
@@ -111,6 +117,10 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
     if (k_lens != nullptr) klimit = min(klimit, k_lens[b]);
     if (q_lens != nullptr) qlimit = min(qlimit, q_lens[b]);
   }
+  const bool has_drop = HAS_EXTRA && p_drop > 0.f;
+  const float inv_keep = HAS_EXTRA ? 1.f / (1.f - p_drop) : 1.f;
+  const unsigned thr24 =
+      HAS_EXTRA ? (unsigned)(p_drop * 16777216.f) : 0u;
 
   // K fragments in registers; V staged once in LDS for the whole workgroup
   bf16x8 kfrag[NT];
@@ -162,6 +172,8 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
 
   for (int gh = 0; gh < gqa; ++gh) {
     const int h = kh * gqa + gh;
+    const float slope =
+        (HAS_EXTRA && alibi != nullptr) ? alibi[h] : 0.f;
     for (int qt = qt0; qt < qt1; ++qt) {
       const int q0 = qt * QT;
       // ---- cooperative staging: natural (swizzled) + transposed tiles ----
@@ -237,9 +249,23 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
         if (HAS_WINDOW && wl >= 0) valid &= (mykey >= qrow + shift - wl);
         if (HAS_WINDOW && wr >= 0 && !CAUSAL)
           valid &= (mykey <= qrow + shift + wr);
-        const float p = valid ? __expf(s[r] * scale - lse_q) : 0.f;
-        s[r] = p;
-        dp[r] = p * (dp[r] - del_q) * scale;
+        float sv = s[r] * scale;
+        if (HAS_EXTRA && alibi != nullptr)
+          sv -= slope * fabsf((float)(qrow + shift - mykey));
+        const float p = valid ? __expf(sv - lse_q) : 0.f;
+        if (HAS_EXTRA && has_drop) {
+          const unsigned long long idx =
+              (((unsigned long long)(b * hq + h) * sq + qrow)) *
+                  (unsigned long long)sk + mykey;
+          const float keep =
+              attn_dropout_keep(rng_seed, idx, thr24) ? inv_keep : 0.f;
+          // dV uses the dropped P; dS = P * (masked dP - delta)
+          s[r] = p * keep;
+          dp[r] = p * (dp[r] * keep - del_q) * scale;
+        } else {
+          s[r] = p;
+          dp[r] = p * (dp[r] - del_q) * scale;
+        }
       }
       // A-frags over the q k-dim; step tp covers q rows 16tp..16tp+15
 #pragma unroll
@@ -291,9 +317,9 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
 // ---------------------------------------------------------------------------
 // 3. dQ kernel (forward structure)
 // ---------------------------------------------------------------------------
-template <int D, bool CAUSAL, bool HAS_WINDOW, bool HAS_LENS>
+template <int D, bool CAUSAL, bool HAS_WINDOW, bool HAS_LENS, bool HAS_EXTRA>
 __global__ __launch_bounds__(512, 2)
-void fa_bwd_dq_kernel(const short* __restrict__ dOut,
+void fa_bwd_dq_kernel_x(const short* __restrict__ dOut,
                       const short* __restrict__ Q,
                       const short* __restrict__ K,
                       const short* __restrict__ V,
@@ -302,7 +328,9 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
                       short* __restrict__ dQ, int b_, int sq, int sk, int hq,
                       int hk, float scale, int wl, int wr,
                       const int* __restrict__ q_lens,
-                      const int* __restrict__ k_lens) {
+                      const int* __restrict__ k_lens,
+                      const float* __restrict__ alibi, float p_drop,
+                      unsigned long long rng_seed) {
   constexpr int NT = D / 16;
   constexpr int NA = D / 32;
   constexpr int KVB = 64;
@@ -354,6 +382,12 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
     }
   }
   const bool row_ok = (qrow < qlimit);
+  const bool has_drop = HAS_EXTRA && p_drop > 0.f;
+  const float inv_keep = HAS_EXTRA ? 1.f / (1.f - p_drop) : 1.f;
+  const unsigned thr24 =
+      HAS_EXTRA ? (unsigned)(p_drop * 16777216.f) : 0u;
+  const float slope =
+      (HAS_EXTRA && alibi != nullptr) ? alibi[h] : 0.f;
   const float lse_q =
       row_ok ? LSE[((long)b * hq + h) * sq + qrow] : INFINITY;
   const float del_q =
@@ -445,8 +479,18 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
           if (HAS_WINDOW && wl >= 0) valid &= (key >= qrow + shift - wl);
           if (HAS_WINDOW && wr >= 0 && !CAUSAL)
             valid &= (key <= qrow + shift + wr);
-          const float p = valid ? __expf(s[kb][r] * scale - lse_q) : 0.f;
-          dp[kb][r] = p * (dp[kb][r] - del_q) * scale;
+          float sv = s[kb][r] * scale;
+          if (HAS_EXTRA && alibi != nullptr)
+            sv -= slope * fabsf((float)(qrow + shift - key));
+          const float p = valid ? __expf(sv - lse_q) : 0.f;
+          float dpv = dp[kb][r];
+          if (HAS_EXTRA && has_drop) {
+            const unsigned long long idx =
+                (((unsigned long long)(b * hq + h) * sq + qrow)) *
+                    (unsigned long long)sk + key;
+            dpv *= attn_dropout_keep(rng_seed, idx, thr24) ? inv_keep : 0.f;
+          }
+          dp[kb][r] = p * (dpv - del_q) * scale;
         }
       }
       // dQ[q][d] += dS[q][key] K[key][d]
@@ -494,13 +538,15 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
 // ---------------------------------------------------------------------------
 
 template <int D>
-static void launch_fa_bwd(const torch::Tensor& dout, const torch::Tensor& q,
+static void launch_fa_bwd_x(const torch::Tensor& dout, const torch::Tensor& q,
                           const torch::Tensor& k, const torch::Tensor& v,
                           const torch::Tensor& lse,
                           const torch::Tensor& delta, torch::Tensor& dq,
                           torch::Tensor& dk, torch::Tensor& dv, float scale,
                           bool causal, int wl, int wr, const int* qlp,
-                          const int* klp, hipStream_t stream) {
+                          const int* klp, const float* alp, float p_drop,
+                          unsigned long long rng_seed,
+                          hipStream_t stream) {
   const int b = q.size(0), sq = q.size(1), hq = q.size(2);
   const int sk = k.size(1), hk = k.size(2);
   const bool has_window = (wl >= 0 || wr >= 0);
@@ -511,45 +557,42 @@ static void launch_fa_bwd(const torch::Tensor& dout, const torch::Tensor& q,
       (const short*)k.data_ptr(), (const short*)v.data_ptr(),                \
       lse.data_ptr<float>(), delta.data_ptr<float>(),                        \
       (short*)dk.data_ptr(), (short*)dv.data_ptr(), b, sq, sk, hq, hk,       \
-      scale, wl, wr, qlp, klp
+      scale, wl, wr, qlp, klp, alp, p_drop, rng_seed
 #define ARGS_DQ                                                              \
   (const short*)dout.data_ptr(), (const short*)q.data_ptr(),                 \
       (const short*)k.data_ptr(), (const short*)v.data_ptr(),                \
       lse.data_ptr<float>(), delta.data_ptr<float>(),                        \
-      (short*)dq.data_ptr(), b, sq, sk, hq, hk, scale, wl, wr, qlp, klp
+      (short*)dq.data_ptr(), b, sq, sk, hq, hk, scale, wl, wr, qlp, klp,    \
+      alp, p_drop, rng_seed
 
   dim3 gkv((sk + 127) / 128, b * hk), bkv(256);
   const int lds_kv = (2 * 32 * D + 2 * D * 32 + 128 * D) * 2 + 2 * 32 * 4;
   dim3 gq((sq + 255) / 256, b * hq), bq(512);
   const int lds_q = 3 * 64 * D * 2;
 
-#define DISPATCH(C, W, L)                                                    \
-  do {                                                                       \
-    hipLaunchKernelGGL((fa_bwd_dkv_kernel<D, C, W, L>), gkv, bkv, lds_kv,    \
-                       stream, ARGS_DKV);                                    \
-    hipLaunchKernelGGL((fa_bwd_dq_kernel<D, C, W, L>), gq, bq, lds_q,        \
+#define DISPATCH_CASE(C, W, L, DR)                                          \
+  if (causal == C && has_window == W && has_lens == L && has_drop == DR) {   \
+    hipLaunchKernelGGL((fa_bwd_dkv_kernel_x<D, C, W, L, DR>), gkv, bkv,        \
+                       lds_kv, stream, ARGS_DKV);                            \
+    hipLaunchKernelGGL((fa_bwd_dq_kernel_x<D, C, W, L, DR>), gq, bq, lds_q,    \
                        stream, ARGS_DQ);                                     \
-  } while (0)
-
-  if (causal) {
-    if (has_window) { if (has_lens) DISPATCH(true, true, true); else DISPATCH(true, true, false); }
-    else { if (has_lens) DISPATCH(true, false, true); else DISPATCH(true, false, false); }
-  } else {
-    if (has_window) { if (has_lens) DISPATCH(false, true, true); else DISPATCH(false, true, false); }
-    else { if (has_lens) DISPATCH(false, false, true); else DISPATCH(false, false, false); }
+    return;                                                                  \
   }
-#undef DISPATCH
+  const bool has_drop = true;  // extra TU: alibi/dropout active
+  DISPATCH_CASE(false, false, false, true)
+  DISPATCH_CASE(false, false, true, true)
+  DISPATCH_CASE(false, true, false, true)
+  DISPATCH_CASE(false, true, true, true)
+  DISPATCH_CASE(true, false, false, true)
+  DISPATCH_CASE(true, false, true, true)
+  DISPATCH_CASE(true, true, false, true)
+  DISPATCH_CASE(true, true, true, true)
+#undef DISPATCH_CASE
 #undef ARGS_DKV
 #undef ARGS_DQ
 }
 
-std::vector<torch::Tensor> fa_backward_extra(
-    torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
-    torch::Tensor out, torch::Tensor lse, double softmax_scale, bool causal,
-    long wl, long wr, torch::Tensor q_lens, torch::Tensor k_lens,
-    torch::Tensor alibi_slopes, double p_drop, long rng_seed);
-
-std::vector<torch::Tensor> fa_backward(torch::Tensor dout, torch::Tensor q,
+std::vector<torch::Tensor> fa_backward_extra(torch::Tensor dout, torch::Tensor q,
                                        torch::Tensor k, torch::Tensor v,
                                        torch::Tensor out, torch::Tensor lse,
                                        double softmax_scale, bool causal,
@@ -557,11 +600,6 @@ std::vector<torch::Tensor> fa_backward(torch::Tensor dout, torch::Tensor q,
                                        torch::Tensor k_lens,
                                        torch::Tensor alibi_slopes,
                                        double p_drop, long rng_seed) {
-  if (alibi_slopes.numel() > 0 || p_drop > 0.0) {
-    return fa_backward_extra(dout, q, k, v, out, lse, softmax_scale, causal,
-                             wl, wr, q_lens, k_lens, alibi_slopes, p_drop,
-                             rng_seed);
-  }
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16,
               "fa_backward: bf16 only");
   TORCH_CHECK(dout.is_contiguous() && q.is_contiguous() &&
@@ -574,7 +612,7 @@ std::vector<torch::Tensor> fa_backward(torch::Tensor dout, torch::Tensor q,
   {
     const long rows = (long)b * sq * hq;
     const long grid = (rows + 3) / 4;
-    hipLaunchKernelGGL(fa_bwd_preprocess_kernel, dim3((unsigned)grid),
+    hipLaunchKernelGGL(fa_bwd_preprocess_kernel_x, dim3((unsigned)grid),
                        dim3(256), 0, stream, (const short*)dout.data_ptr(),
                        (const short*)out.data_ptr(), delta.data_ptr<float>(),
                        b, sq, hq, D);
@@ -586,14 +624,20 @@ std::vector<torch::Tensor> fa_backward(torch::Tensor dout, torch::Tensor q,
   auto kl = k_lens.numel() ? k_lens.to(q.device(), torch::kInt32) : k_lens;
   const int* qlp = ql.numel() ? ql.data_ptr<int>() : nullptr;
   const int* klp = kl.numel() ? kl.data_ptr<int>() : nullptr;
+  auto al = alibi_slopes.numel()
+                ? alibi_slopes.to(q.device(), torch::kFloat32).contiguous()
+                : alibi_slopes;
+  const float* alp = al.numel() ? al.data_ptr<float>() : nullptr;
   if (D == 128) {
-    launch_fa_bwd<128>(dout, q, k, v, lse_c, delta, dq, dk, dv,
+    launch_fa_bwd_x<128>(dout, q, k, v, lse_c, delta, dq, dk, dv,
                        (float)softmax_scale, causal, (int)wl, (int)wr, qlp,
-                       klp, stream);
+                       klp, alp, (float)p_drop,
+                       (unsigned long long)rng_seed, stream);
   } else {
-    launch_fa_bwd<64>(dout, q, k, v, lse_c, delta, dq, dk, dv,
+    launch_fa_bwd_x<64>(dout, q, k, v, lse_c, delta, dq, dk, dv,
                       (float)softmax_scale, causal, (int)wl, (int)wr, qlp,
-                      klp, stream);
+                      klp, alp, (float)p_drop,
+                      (unsigned long long)rng_seed, stream);
   }
   HIP_CHECK_LAST();
   return {dq, dk, dv};

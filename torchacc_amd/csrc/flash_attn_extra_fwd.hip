@@ -1,4 +1,7 @@
-#include "hip/hip_runtime.h"
+// Full-featured flash-attention FORWARD (alibi + dropout), separate
+// translation unit: the vanilla hot path (flash_attn_fwd.hip) stays at its
+// tuned register budget; these variants tolerate spills (guide rule 19:
+// co-compiled template variants perturb each other's codegen).
 // Flash-attention v2 FORWARD for gfx950 (CDNA4 MFMA).
 //
 // Structure (per the CDNA4 attention ladder in the MI355X guides):
@@ -20,30 +23,23 @@
 // per-batch varlen (q_lens/k_lens), D in {64, 128}. bf16 only.
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
-#include "common.h"
-
-#define LOG2E 1.4426950408889634f
+#include "attn_common.h"
 
 typedef float f32x16_ __attribute__((ext_vector_type(16)));
+#define cvt_pk_bf16 attn_cvt_pk_bf16
 
-DEVINLINE unsigned cvt_pk_bf16(float lo, float hi) {
-  unsigned r;
-  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
-               : "=v"(r) : "v"(lo), "v"(hi));
-  return r;
-}
-
-// crow: D-matrix row for accumulator register r on lane-half hi (32x32 mfma)
-#define CROW(r, hi) (((r) & 3) + 8 * ((r) >> 2) + 4 * (hi))
-
-template <int D, bool CAUSAL, bool HAS_WINDOW, bool HAS_LENS>
+// HAS_EXTRA: alibi and/or dropout active (kept out of the
+// register budget of the vanilla path)
+template <int D, bool CAUSAL, bool HAS_WINDOW, bool HAS_LENS, bool HAS_EXTRA>
 __global__ __launch_bounds__(512, 2)
-void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
+void fa_fwd_kernel_x(const short* __restrict__ Q, const short* __restrict__ K,
                    const short* __restrict__ V, short* __restrict__ O,
                    float* __restrict__ LSE, int b_, int sq, int sk, int hq,
                    int hk, float scale, int wl, int wr,
                    const int* __restrict__ q_lens,
-                   const int* __restrict__ k_lens) {
+                   const int* __restrict__ k_lens,
+                   const float* __restrict__ alibi, float p_drop,
+                   unsigned long long rng_seed) {
   constexpr int NT = D / 16;   // QK^T k-steps (d slices of 16)
   constexpr int NA = D / 32;   // PV output accs (d blocks of 32)
   constexpr int KVB = 64;      // keys per tile
@@ -79,6 +75,13 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
     if (k_lens != nullptr) klimit = min(klimit, k_lens[b]);
     if (q_lens != nullptr) qlimit = min(qlimit, q_lens[b]);
   }
+  const float slope =
+      (HAS_EXTRA && alibi != nullptr) ? alibi[h] : 0.f;
+  const bool has_alibi = HAS_EXTRA && alibi != nullptr;
+  const bool has_drop = HAS_EXTRA && p_drop > 0.f;
+  const float inv_keep = HAS_EXTRA ? 1.f / (1.f - p_drop) : 1.f;
+  const unsigned thr24 =
+      HAS_EXTRA ? (unsigned)(p_drop * 16777216.f) : 0u;
 
   // ---- load Q fragments (pre-scaled) -------------------------------------
   // B-frag for swapped QK^T: lane holds Q[qrow][t*16 + hi*8 + j], j=0..7
@@ -208,7 +211,7 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
       // "full" = no per-element masking needed for ANY q row of this wave:
       // causal bound uses the wave's SMALLEST q (q0), window-left bound its
       // LARGEST q (q0+31), window-right its smallest.
-      const bool tile_full =
+      const bool tile_full = !has_alibi &&
           (!CAUSAL || kv0 + KVB - 1 <= q0 + shift) &&
           (kv0 + KVB <= klimit) &&
           (!HAS_WINDOW || wl < 0 || kv0 >= q0 + 31 + shift - wl) &&
@@ -222,6 +225,8 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
           float s = p[kb][r] * scale;
           if (!tile_full) {
             const int key = kv0 + kb * 32 + CROW(r, hi);
+            if (HAS_EXTRA && has_alibi)
+              s -= slope * fabsf((float)(qrow + shift - key));
             bool valid = key < klimit;
             if (CAUSAL) valid &= (key <= qrow + shift);
             if (HAS_WINDOW && wl >= 0) valid &= (key >= qrow + shift - wl);
@@ -252,6 +257,21 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
       }
       lsum += __shfl_xor(lsum, 32, 64);
       l_run = l_run * alpha + lsum;
+      if (HAS_EXTRA && has_drop) {
+        // dropout AFTER the softmax statistics: lse is unaffected
+#pragma unroll
+        for (int kb = 0; kb < 2; ++kb) {
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int key = kv0 + kb * 32 + CROW(r, hi);
+            const unsigned long long idx =
+                (((unsigned long long)(b * hq + h) * sq + qrow)) *
+                    (unsigned long long)sk + key;
+            p[kb][r] = attn_dropout_keep(rng_seed, idx, thr24)
+                           ? p[kb][r] * inv_keep : 0.f;
+          }
+        }
+      }
 #pragma unroll
       for (int a = 0; a < NA; ++a) {
 #pragma unroll
@@ -339,11 +359,13 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
 // ---------------------------------------------------------------------------
 
 template <int D>
-static void launch_fa_fwd(const torch::Tensor& q, const torch::Tensor& k,
+static void launch_fa_fwd_x(const torch::Tensor& q, const torch::Tensor& k,
                           const torch::Tensor& v, torch::Tensor& o,
                           torch::Tensor& lse, float scale, bool causal,
                           int wl, int wr, const torch::Tensor& q_lens,
-                          const torch::Tensor& k_lens, hipStream_t stream) {
+                          const torch::Tensor& k_lens,
+                          const torch::Tensor& alibi, float p_drop,
+                          unsigned long long rng_seed, hipStream_t stream) {
   const int b = q.size(0), sq = q.size(1), hq = q.size(2);
   const int sk = k.size(1), hk = k.size(2);
   const bool has_window = (wl >= 0 || wr >= 0);
@@ -353,42 +375,37 @@ static void launch_fa_fwd(const torch::Tensor& q, const torch::Tensor& k,
   const int lds = 4 * 64 * D * 2;  // K + V^T, double buffered
   const int* qlp = q_lens.numel() ? q_lens.data_ptr<int>() : nullptr;
   const int* klp = k_lens.numel() ? k_lens.data_ptr<int>() : nullptr;
+  const float* alp = alibi.numel() ? alibi.data_ptr<float>() : nullptr;
 
-#define LAUNCH(CAUSAL, WIN, LENS)                                            \
-  hipLaunchKernelGGL((fa_fwd_kernel<D, CAUSAL, WIN, LENS>), grid, block,     \
-                     lds, stream, (const short*)q.data_ptr(),                \
-                     (const short*)k.data_ptr(),                             \
-                     (const short*)v.data_ptr(), (short*)o.data_ptr(),       \
-                     lse.data_ptr<float>(), b, sq, sk, hq, hk, scale, wl,    \
-                     wr, qlp, klp)
-  if (causal) {
-    if (has_window) { if (has_lens) LAUNCH(true, true, true); else LAUNCH(true, true, false); }
-    else { if (has_lens) LAUNCH(true, false, true); else LAUNCH(true, false, false); }
-  } else {
-    if (has_window) { if (has_lens) LAUNCH(false, true, true); else LAUNCH(false, true, false); }
-    else { if (has_lens) LAUNCH(false, false, true); else LAUNCH(false, false, false); }
+#define LAUNCH_CASE(C, W, L, DR)                                            \
+  if (causal == C && has_window == W && has_lens == L && has_drop == DR) {   \
+    hipLaunchKernelGGL((fa_fwd_kernel_x<D, C, W, L, DR>), grid, block, lds,    \
+                       stream, (const short*)q.data_ptr(),                   \
+                       (const short*)k.data_ptr(),                           \
+                       (const short*)v.data_ptr(), (short*)o.data_ptr(),     \
+                       lse.data_ptr<float>(), b, sq, sk, hq, hk, scale, wl,  \
+                       wr, qlp, klp, alp, p_drop, rng_seed);                 \
+    return;                                                                  \
   }
-#undef LAUNCH
+  const bool has_drop = true;  // extra TU: alibi/dropout active
+  LAUNCH_CASE(false, false, false, true)
+  LAUNCH_CASE(false, false, true, true)
+  LAUNCH_CASE(false, true, false, true)
+  LAUNCH_CASE(false, true, true, true)
+  LAUNCH_CASE(true, false, false, true)
+  LAUNCH_CASE(true, false, true, true)
+  LAUNCH_CASE(true, true, false, true)
+  LAUNCH_CASE(true, true, true, true)
+#undef LAUNCH_CASE
 }
 
-// full-featured variants (alibi + dropout), separate TU
-std::vector<torch::Tensor> fa_forward_extra(
-    torch::Tensor q, torch::Tensor k, torch::Tensor v, double softmax_scale,
-    bool causal, long wl, long wr, torch::Tensor q_lens,
-    torch::Tensor k_lens, torch::Tensor alibi_slopes, double p_drop,
-    long rng_seed);
-
-std::vector<torch::Tensor> fa_forward(torch::Tensor q, torch::Tensor k,
+std::vector<torch::Tensor> fa_forward_extra(torch::Tensor q, torch::Tensor k,
                                       torch::Tensor v, double softmax_scale,
                                       bool causal, long wl, long wr,
                                       torch::Tensor q_lens,
                                       torch::Tensor k_lens,
                                       torch::Tensor alibi_slopes,
                                       double p_drop, long rng_seed) {
-  if (alibi_slopes.numel() > 0 || p_drop > 0.0) {
-    return fa_forward_extra(q, k, v, softmax_scale, causal, wl, wr, q_lens,
-                            k_lens, alibi_slopes, p_drop, rng_seed);
-  }
   TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
               v.is_contiguous());
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16,
@@ -403,13 +420,20 @@ std::vector<torch::Tensor> fa_forward(torch::Tensor q, torch::Tensor k,
                            : q_lens;
   auto kl = k_lens.numel() ? k_lens.to(q.device(), torch::kInt32)
                            : k_lens;
+  auto al = alibi_slopes.numel()
+                ? alibi_slopes.to(q.device(), torch::kFloat32).contiguous()
+                : alibi_slopes;
+  TORCH_CHECK(al.numel() == 0 || al.numel() == q.size(2),
+              "alibi_slopes must be [num_heads]");
   auto stream = at::hip::getCurrentHIPStream();
   if (D == 128) {
-    launch_fa_fwd<128>(q, k, v, o, lse, (float)softmax_scale, causal,
-                       (int)wl, (int)wr, ql, kl, stream);
+    launch_fa_fwd_x<128>(q, k, v, o, lse, (float)softmax_scale, causal,
+                       (int)wl, (int)wr, ql, kl, al, (float)p_drop,
+                       (unsigned long long)rng_seed, stream);
   } else {
-    launch_fa_fwd<64>(q, k, v, o, lse, (float)softmax_scale, causal, (int)wl,
-                      (int)wr, ql, kl, stream);
+    launch_fa_fwd_x<64>(q, k, v, o, lse, (float)softmax_scale, causal, (int)wl,
+                      (int)wr, ql, kl, al, (float)p_drop,
+                      (unsigned long long)rng_seed, stream);
   }
   HIP_CHECK_LAST();
   return {o, lse};
