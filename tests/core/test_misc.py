@@ -109,3 +109,20 @@ def test_hf_kernel_patches_apply():
     ref = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) +
                           m.variance_epsilon) * m.weight
     assert torch.allclose(y, ref, atol=1e-5)
+
+
+def test_qwen2_native_trains():
+    from torchacc_amd.models import Qwen2ForCausalLM, qwen2_tiny
+    torch.manual_seed(0)
+    model = Qwen2ForCausalLM(qwen2_tiny(sliding_window=16))
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
+    torch.manual_seed(1)
+    ids = torch.randint(0, 1024, (2, 64))
+    losses = []
+    for _ in range(5):
+        loss = model(ids, labels=ids)
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0], losses
