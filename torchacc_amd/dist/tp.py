@@ -90,29 +90,15 @@ class ColumnParallelLinear(nn.Module):
         nn.init.kaiming_uniform_(self.weight, a=math.sqrt(5))
 
     @classmethod
-    def from_linear(cls, lin: nn.Linear, group,
-                    sections=None) -> "ColumnParallelLinear":
-        """``sections``: for merged projections (qkv / gate_up), shard each
-        section independently and concatenate the per-rank slices so the
-        local layout stays [q_r; k_r; v_r]."""
+    def from_linear(cls, lin: nn.Linear, group) -> "ColumnParallelLinear":
         m = cls(lin.in_features, lin.out_features, group,
                 bias=lin.bias is not None, dtype=lin.weight.dtype,
                 device=lin.weight.device)
         rank = dist.get_rank(group)
         with torch.no_grad():
-            if sections is None:
-                m.weight.copy_(lin.weight.chunk(m.tp, dim=0)[rank])
-                if lin.bias is not None:
-                    m.bias.copy_(lin.bias.chunk(m.tp, dim=0)[rank])
-            else:
-                wparts = lin.weight.split(list(sections), dim=0)
-                m.weight.copy_(torch.cat(
-                    [p.chunk(m.tp, dim=0)[rank] for p in wparts], dim=0))
-                if lin.bias is not None:
-                    bparts = lin.bias.split(list(sections), dim=0)
-                    m.bias.copy_(torch.cat(
-                        [p.chunk(m.tp, dim=0)[rank] for p in bparts],
-                        dim=0))
+            m.weight.copy_(lin.weight.chunk(m.tp, dim=0)[rank])
+            if lin.bias is not None:
+                m.bias.copy_(lin.bias.chunk(m.tp, dim=0)[rank])
         return m
 
     def forward(self, x):
@@ -218,28 +204,26 @@ def parallelize_module(model: nn.Module, config) -> nn.Module:
     if tp == 1 or group is None:
         return model
     from ..models.llama import LlamaAttention, LlamaMLP
-    from ..models.qwen2 import Qwen2Attention
     n_attn = n_mlp = 0
     for mod in model.modules():
-        if isinstance(mod, (LlamaAttention, Qwen2Attention)):
+        if isinstance(mod, LlamaAttention):
             assert mod.num_heads % tp == 0, \
                 f"attention heads {mod.num_heads} not divisible by tp {tp}"
             assert mod.num_kv_heads % tp == 0, \
                 (f"kv heads {mod.num_kv_heads} not divisible by tp {tp}; "
                  "use a smaller tp degree for this GQA config")
-            hd = mod.head_dim
-            sections = (mod.num_heads * hd, mod.num_kv_heads * hd,
-                        mod.num_kv_heads * hd)
-            mod.qkv_proj = ColumnParallelLinear.from_linear(
-                mod.qkv_proj, group, sections=sections)
+            mod.q_proj = ColumnParallelLinear.from_linear(mod.q_proj, group)
+            mod.k_proj = ColumnParallelLinear.from_linear(mod.k_proj, group)
+            mod.v_proj = ColumnParallelLinear.from_linear(mod.v_proj, group)
             mod.o_proj = RowParallelLinear.from_linear(mod.o_proj, group)
             mod.num_heads //= tp
             mod.num_kv_heads //= tp
             n_attn += 1
         elif isinstance(mod, LlamaMLP):
-            half = mod.gate_up_proj.out_features // 2
-            mod.gate_up_proj = ColumnParallelLinear.from_linear(
-                mod.gate_up_proj, group, sections=(half, half))
+            mod.gate_proj = ColumnParallelLinear.from_linear(
+                mod.gate_proj, group)
+            mod.up_proj = ColumnParallelLinear.from_linear(mod.up_proj,
+                                                           group)
             mod.down_proj = RowParallelLinear.from_linear(mod.down_proj,
                                                           group)
             n_mlp += 1
