@@ -90,15 +90,29 @@ class ColumnParallelLinear(nn.Module):
         nn.init.kaiming_uniform_(self.weight, a=math.sqrt(5))
 
     @classmethod
-    def from_linear(cls, lin: nn.Linear, group) -> "ColumnParallelLinear":
+    def from_linear(cls, lin: nn.Linear, group,
+                    sections=None) -> "ColumnParallelLinear":
+        """``sections``: for merged projections (qkv / gate_up), shard each
+        section independently and concatenate the per-rank slices so the
+        local layout stays [q_r; k_r; v_r]."""
         m = cls(lin.in_features, lin.out_features, group,
                 bias=lin.bias is not None, dtype=lin.weight.dtype,
                 device=lin.weight.device)
         rank = dist.get_rank(group)
         with torch.no_grad():
-            m.weight.copy_(lin.weight.chunk(m.tp, dim=0)[rank])
-            if lin.bias is not None:
-                m.bias.copy_(lin.bias.chunk(m.tp, dim=0)[rank])
+            if sections is None:
+                m.weight.copy_(lin.weight.chunk(m.tp, dim=0)[rank])
+                if lin.bias is not None:
+                    m.bias.copy_(lin.bias.chunk(m.tp, dim=0)[rank])
+            else:
+                wparts = lin.weight.split(list(sections), dim=0)
+                m.weight.copy_(torch.cat(
+                    [p.chunk(m.tp, dim=0)[rank] for p in wparts], dim=0))
+                if lin.bias is not None:
+                    bparts = lin.bias.split(list(sections), dim=0)
+                    m.bias.copy_(torch.cat(
+                        [p.chunk(m.tp, dim=0)[rank] for p in bparts],
+                        dim=0))
         return m
 
     def forward(self, x):
@@ -204,9 +218,10 @@ def parallelize_module(model: nn.Module, config) -> nn.Module:
     if tp == 1 or group is None:
         return model
     from ..models.llama import LlamaAttention, LlamaMLP
+    from ..models.qwen2 import Qwen2Attention
     n_attn = n_mlp = 0
     for mod in model.modules():
-        if isinstance(mod, LlamaAttention):
+        if isinstance(mod, (LlamaAttention, Qwen2Attention)):
             assert mod.num_heads % tp == 0, \
                 f"attention heads {mod.num_heads} not divisible by tp {tp}"
             assert mod.num_kv_heads % tp == 0, \

@@ -59,7 +59,6 @@ def llama_tiny(**kw) -> "LlamaConfig":
 
 
 class LlamaAttention(nn.Module):
-    """QKV fused into one projection GEMM (fewer, larger hipBLASLt calls)."""
 
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
@@ -69,20 +68,21 @@ class LlamaAttention(nn.Module):
         self.num_heads = h
         self.num_kv_heads = hk
         self.head_dim = cfg.hidden_size // h
-        self.qkv_proj = nn.Linear(
-            cfg.hidden_size, (h + 2 * hk) * self.head_dim, bias=False)
+        self.q_proj = nn.Linear(cfg.hidden_size, h * self.head_dim,
+                                bias=False)
+        self.k_proj = nn.Linear(cfg.hidden_size, hk * self.head_dim,
+                                bias=False)
+        self.v_proj = nn.Linear(cfg.hidden_size, hk * self.head_dim,
+                                bias=False)
         self.o_proj = nn.Linear(h * self.head_dim, cfg.hidden_size,
                                 bias=False)
 
     def forward(self, x, cos, sin):
         b, s, _ = x.shape
         h, hk = self.num_heads, self.num_kv_heads
-        qkv = self.qkv_proj(x)
-        q, k, v = qkv.split([h * self.head_dim, hk * self.head_dim,
-                             hk * self.head_dim], dim=-1)
-        q = q.view(b, s, h, self.head_dim)
-        k = k.contiguous().view(b, s, hk, self.head_dim)
-        v = v.contiguous().view(b, s, hk, self.head_dim)
+        q = self.q_proj(x).view(b, s, h, self.head_dim)
+        k = self.k_proj(x).view(b, s, hk, self.head_dim)
+        v = self.v_proj(x).view(b, s, hk, self.head_dim)
         mode = self.cfg.cp_mode
         if mode is None:
             q, k = apply_rotary_pos_emb(q, k, cos, sin)
@@ -118,20 +118,18 @@ class LlamaAttention(nn.Module):
 
 
 class LlamaMLP(nn.Module):
-    """gate and up fused into one projection GEMM."""
 
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
-        self.intermediate_size = cfg.intermediate_size
-        self.gate_up_proj = nn.Linear(
-            cfg.hidden_size, 2 * cfg.intermediate_size, bias=False)
+        self.gate_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size,
+                                   bias=False)
+        self.up_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size,
+                                 bias=False)
         self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size,
                                    bias=False)
 
     def forward(self, x):
-        gu = self.gate_up_proj(x)
-        g, u = gu.chunk(2, dim=-1)
-        return self.down_proj(swiglu(g.contiguous(), u.contiguous()))
+        return self.down_proj(swiglu(self.gate_proj(x), self.up_proj(x)))
 
 
 class LlamaDecoderLayer(nn.Module):

@@ -53,20 +53,20 @@ class Qwen2Attention(nn.Module):
         self.num_heads = h
         self.num_kv_heads = hk
         self.head_dim = cfg.hidden_size // h
-        self.qkv_proj = nn.Linear(
-            cfg.hidden_size, (h + 2 * hk) * self.head_dim, bias=True)
+        self.q_proj = nn.Linear(cfg.hidden_size, h * self.head_dim, bias=True)
+        self.k_proj = nn.Linear(cfg.hidden_size, hk * self.head_dim,
+                                bias=True)
+        self.v_proj = nn.Linear(cfg.hidden_size, hk * self.head_dim,
+                                bias=True)
         self.o_proj = nn.Linear(h * self.head_dim, cfg.hidden_size,
                                 bias=False)
 
     def forward(self, x, cos, sin):
         b, s, _ = x.shape
         h, hk = self.num_heads, self.num_kv_heads
-        qkv = self.qkv_proj(x)
-        q, k, v = qkv.split([h * self.head_dim, hk * self.head_dim,
-                             hk * self.head_dim], dim=-1)
-        q = q.view(b, s, h, self.head_dim)
-        k = k.contiguous().view(b, s, hk, self.head_dim)
-        v = v.contiguous().view(b, s, hk, self.head_dim)
+        q = self.q_proj(x).view(b, s, h, self.head_dim)
+        k = self.k_proj(x).view(b, s, hk, self.head_dim)
+        v = self.v_proj(x).view(b, s, hk, self.head_dim)
         q, k = apply_rotary_pos_emb(q, k, cos, sin)
         window = (-1, -1)
         if self.cfg.sliding_window is not None and \
