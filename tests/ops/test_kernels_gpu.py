@@ -364,3 +364,23 @@ def test_fa_dropout():
                                  e, e, e, 0.2, 7)
     for t in (dq, dk, dv):
         assert torch.isfinite(t.float()).all()
+
+
+def test_fa_varlen_position_ids():
+    """Packed-sequence path (HF packed-sample training)."""
+    from torchacc_amd.ops.flash_attn import (
+        flash_attn_varlen_position_ids_xla)
+    torch.manual_seed(0)
+    lens = [96, 160, 64]
+    total = sum(lens)
+    pos = torch.cat([torch.arange(n) for n in lens]).unsqueeze(0)
+    h, d = 4, 128
+    q = torch.randn(1, total, h, d)
+    k = torch.randn(1, total, h, d)
+    v = torch.randn(1, total, h, d)
+    ref = flash_attn_varlen_position_ids_xla(q, k, v, pos, causal=True)
+    qg, kg, vg = _to_gpu(q, k, v)
+    out = flash_attn_varlen_position_ids_xla(qg, kg, vg, pos.cuda(),
+                                             causal=True)
+    err = (out.float().cpu() - ref.float()).abs().max()
+    assert err < 2.5e-2, err
