@@ -94,6 +94,8 @@ class FlashAttnFunc(torch.autograd.Function):
             softmax_scale = 1.0 / math.sqrt(q.shape[-1])
         q, k, v = [t.contiguous() for t in (q, k, v)]
         ext = dispatch(q)
+        if q.dtype != torch.bfloat16:
+            ext = None  # CDNA4 attention kernels are bf16
         wl, wr = window_size
         al = alibi_slopes if alibi_slopes is not None else torch.empty(0)
         seed = 0
@@ -130,6 +132,8 @@ class FlashAttnFunc(torch.autograd.Function):
         q_lens = q_lens if q_lens.numel() else None
         k_lens = k_lens if k_lens.numel() else None
         ext = dispatch(q)
+        if q.dtype != torch.bfloat16:
+            ext = None
         dout = dout.contiguous()
         wl, wr = ctx.window
         if ext is not None:

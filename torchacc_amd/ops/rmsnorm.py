@@ -21,6 +21,8 @@ class _RMSNorm(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, eps):
         ext = dispatch(x)
+        if x.dtype not in (torch.bfloat16, torch.float32):
+            ext = None
         x = x.contiguous()
         if ext is not None:
             y, inv_rms = ext.rmsnorm_forward(x, weight, eps)
@@ -35,6 +37,8 @@ class _RMSNorm(torch.autograd.Function):
     def backward(ctx, dy):
         x, weight, inv_rms = ctx.saved_tensors
         ext = dispatch(x)
+        if x.dtype != torch.bfloat16:
+            ext = None  # kernel backward is bf16-only
         dy = dy.contiguous()
         if ext is not None:
             dx, dw = ext.rmsnorm_backward(dy, x, weight, inv_rms)
@@ -79,6 +83,8 @@ class _AddRMSNorm(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, residual, weight, eps):
         ext = dispatch(x)
+        if x.dtype != torch.bfloat16:
+            ext = None  # composite path for fp32-on-GPU
         x = x.contiguous()
         has_resid = residual is not None
         if ext is not None:
@@ -99,6 +105,8 @@ class _AddRMSNorm(torch.autograd.Function):
     def backward(ctx, dy, dresid):
         r, weight, inv_rms = ctx.saved_tensors
         ext = dispatch(r)
+        if r.dtype != torch.bfloat16:
+            ext = None
         dy = dy.contiguous()
         if ext is not None:
             dx, dw = ext.add_rmsnorm_backward(

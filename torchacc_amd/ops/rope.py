@@ -44,6 +44,8 @@ class _RoPE(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, cos, sin):
         ext = dispatch(q)
+        if q.dtype != torch.bfloat16:
+            ext = None
         q = q.contiguous()
         k = k.contiguous()
         cos = cos.float()
@@ -61,6 +63,8 @@ class _RoPE(torch.autograd.Function):
         cos = cos.float()
         sin = sin.float()
         ext = dispatch(dq)
+        if dq.dtype != torch.bfloat16:
+            ext = None
         dq = dq.contiguous()
         dk = dk.contiguous()
         if ext is not None:

@@ -13,6 +13,8 @@ class _SwiGLU(torch.autograd.Function):
     @staticmethod
     def forward(ctx, gate, up):
         ext = dispatch(gate)
+        if gate.dtype != torch.bfloat16:
+            ext = None
         gate = gate.contiguous()
         up = up.contiguous()
         ctx.save_for_backward(gate, up)
@@ -25,6 +27,8 @@ class _SwiGLU(torch.autograd.Function):
     def backward(ctx, dy):
         gate, up = ctx.saved_tensors
         ext = dispatch(gate)
+        if gate.dtype != torch.bfloat16:
+            ext = None
         dy = dy.contiguous()
         if ext is not None:
             dgate, dup = ext.swiglu_backward(dy, gate, up)
