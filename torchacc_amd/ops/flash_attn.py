@@ -28,8 +28,8 @@ from ._backend import dispatch
 
 def _check_qkv(q, k, v):
     if q.is_cuda:
-        assert q.dtype in (torch.float16, torch.bfloat16), \
-            "flash attention requires fp16/bf16"
+        assert q.dtype in (torch.float16, torch.bfloat16, torch.float32), \
+            "flash attention requires fp16/bf16 (or fp32 composite)"
     assert q.dtype == k.dtype == v.dtype
     assert q.dim() == 4 and k.dim() == 4 and v.dim() == 4, \
         "expected [batch, seqlen, heads, head_dim]"
