@@ -43,13 +43,13 @@ def main():
     causal = args.causal
     flops_fwd = 4 * b * h * s * s * d * (0.5 if causal else 1.0)
 
-    out, lse = ext.fa_forward(q, k, v, scale, causal, -1, -1,
-                              torch.empty(0), torch.empty(0))
+    e = torch.empty(0)
+    out, lse = ext.fa_forward(q, k, v, scale, causal, -1, -1, e, e, e,
+                              0.0, 0)
     t_fwd = bench(lambda: ext.fa_forward(q, k, v, scale, causal, -1, -1,
-                                         torch.empty(0), torch.empty(0)))
+                                         e, e, e, 0.0, 0))
     t_bwd = bench(lambda: ext.fa_backward(do, q, k, v, out, lse, scale,
-                                          causal, -1, -1, torch.empty(0),
-                                          torch.empty(0)))
+                                          causal, -1, -1, e, e, e, 0.0, 0))
     print(json.dumps({
         "shape": f"b{b} s{s} h{h} hk{hk} d{d} causal={causal}",
         "fwd_ms": t_fwd * 1e3,
