@@ -113,6 +113,7 @@ class FlatParamUnit:
             setattr(mod, attr, None)
         self._views_valid = False
         self.in_backward = False
+        self.pending_bwd = 0   # grad-enabled forwards awaiting backward
         self.unsharded = True  # storage currently allocated (init state)
         self.ag_event: Optional[torch.cuda.Event] = None
         self._post_bwd_hooked = False
@@ -350,6 +351,8 @@ class FullyShardedDataParallel(ParallelModule):
         unit.wait_unshard()
         track = torch.is_grad_enabled()
         unit.rebuild_views(track_grad=track)
+        if track:
+            unit.pending_bwd += 1
         if track and not unit._post_bwd_hooked:
             unit.full_flat.register_post_accumulate_grad_hook(
                 functools.partial(self._post_backward_unit, unit))
@@ -402,7 +405,11 @@ class FullyShardedDataParallel(ParallelModule):
     def _post_backward_unit(self, unit: FlatParamUnit, _leaf):
         unit.reduce_grad(self.grad_scale, self.dp_group, self._rs_stream())
         unit.in_backward = False
-        unit.reshard()
+        unit.pending_bwd = max(0, unit.pending_bwd - 1)
+        # pipeline micro-batches: later backwards still hold saved views
+        # into this storage — reshard only after the last one
+        if unit.pending_bwd == 0:
+            unit.reshard()
 
     # ---- nn.Module API ---------------------------------------------------
 
