@@ -85,3 +85,34 @@ def test_fsdp_full_state_dict_roundtrip():
     for _ in range(2):
         rank, ok, sample = q.get()
         assert ok, f"rank {rank} full_state_dict mismatch ({sample})"
+
+
+def _gc_reshard_worker(rank, world, q):
+    import torchacc_amd as ta
+    cfg = ta.Config()
+    cfg.dist.fsdp.size = world
+    cfg.dist.fsdp.wrap_layer_cls = {"LlamaDecoderLayer"}
+    cfg.memory.gc = True
+    cfg.memory.gc_cls = {"LlamaDecoderLayer"}
+    model = _make_model()
+    wrapped = ta.accelerate(model, config=cfg)
+    fsdp = wrapped.fsdp_wrapper
+    torch.manual_seed(3)
+    ids = torch.randint(0, 1024, (2, 32))
+    for _ in range(2):
+        loss = wrapped(ids, labels=ids)
+        loss.backward()
+    # after backward every non-root unit must be resharded (pending == 0)
+    bad = [u.name for u in fsdp.units
+           if u is not fsdp.root_unit and (u.unsharded or u.pending_bwd)]
+    q.put((rank, bad))
+
+
+def test_gc_units_reshard_after_backward():
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    run_multiprocess(_gc_reshard_worker, world_size=2, args=(q,))
+    for _ in range(2):
+        rank, bad = q.get()
+        assert not bad, f"rank {rank}: units left gathered {bad}"

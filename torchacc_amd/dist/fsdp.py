@@ -351,7 +351,9 @@ class FullyShardedDataParallel(ParallelModule):
         unit.wait_unshard()
         track = torch.is_grad_enabled()
         unit.rebuild_views(track_grad=track)
-        if track:
+        if track and not unit.in_backward:
+            # (a gradient-checkpoint RECOMPUTE re-enters this hook during
+            # backward; it must not count as a new pending backward)
             unit.pending_bwd += 1
         if track and not unit._post_bwd_hooked:
             unit.full_flat.register_post_accumulate_grad_hook(
