@@ -359,7 +359,11 @@ class FullyShardedDataParallel(ParallelModule):
             unit.full_flat.register_post_accumulate_grad_hook(
                 functools.partial(self._post_backward_unit, unit))
             unit._post_bwd_hooked = True
-        self._prefetch_next(unit, +1)
+        if not unit.in_backward:
+            # (during a gradient-checkpoint recompute the next unit's
+            # backward has already finished — prefetching it would leave it
+            # gathered forever)
+            self._prefetch_next(unit, +1)
         return None
 
     def _post_forward_unit(self, unit, mod, args, output):
