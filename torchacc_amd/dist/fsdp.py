@@ -116,6 +116,7 @@ class FlatParamUnit:
         self.pending_bwd = 0   # grad-enabled forwards awaiting backward
         self.unsharded = True  # storage currently allocated (init state)
         self.ag_event: Optional[torch.cuda.Event] = None
+        self.rs_event: Optional[torch.cuda.Event] = None
         self._post_bwd_hooked = False
 
     # ---- unshard / reshard ---------------------------------------------
@@ -180,44 +181,54 @@ class FlatParamUnit:
     # ---- gradient path --------------------------------------------------
 
     def reduce_grad(self, grad_scale: float, dp_group, rs_stream):
-        """reduce-scatter full grad -> shard grad (+ hybrid DP all-reduce)."""
+        """reduce-scatter full grad -> shard grad (+ hybrid DP all-reduce).
+
+        On GPU the whole reduction (and the accumulate into shard.grad) runs
+        on the side stream so it OVERLAPS the rest of backward; the compute
+        stream only syncs once, at the end of backward
+        (FullyShardedDataParallel._finalize_grad_sync). The per-unit event
+        is recorded in ``rs_event``."""
         full_grad = self.full_flat.grad
         if full_grad is None:
             return
-        if self.ws > 1:
-            out = torch.empty(self.shard_numel, dtype=full_grad.dtype,
-                              device=full_grad.device)
-            if rs_stream is not None:
-                rs_stream.wait_stream(torch.cuda.current_stream())
-                with torch.cuda.stream(rs_stream):
+        if rs_stream is not None:
+            rs_stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(rs_stream):
+                if self.ws > 1:
+                    out = torch.empty(self.shard_numel,
+                                      dtype=full_grad.dtype,
+                                      device=full_grad.device)
                     dist.reduce_scatter_tensor(out, full_grad,
                                                group=self.group)
-                    if dp_group is not None:
-                        dist.all_reduce(out, group=dp_group)
-                    full_grad.record_stream(rs_stream)
-                torch.cuda.current_stream().wait_stream(rs_stream)
-            else:
-                dist.reduce_scatter_tensor(out, full_grad, group=self.group)
+                else:
+                    out = full_grad
                 if dp_group is not None:
                     dist.all_reduce(out, group=dp_group)
-        else:
-            out = full_grad
-            if dp_group is not None:
-                if rs_stream is not None:
-                    rs_stream.wait_stream(torch.cuda.current_stream())
-                    with torch.cuda.stream(rs_stream):
-                        dist.all_reduce(out, group=dp_group)
-                        full_grad.record_stream(rs_stream)
-                    torch.cuda.current_stream().wait_stream(rs_stream)
+                if grad_scale != 1.0:
+                    out.mul_(grad_scale)
+                if self.shard.grad is None:
+                    self.shard.grad = out
                 else:
-                    dist.all_reduce(out, group=dp_group)
-            # no clone: next backward allocates a fresh full_flat.grad
-        if grad_scale != 1.0:
-            out.mul_(grad_scale)
-        if self.shard.grad is None:
-            self.shard.grad = out
+                    self.shard.grad.add_(out)
+                full_grad.record_stream(rs_stream)
+            ev = torch.cuda.Event()
+            ev.record(rs_stream)
+            self.rs_event = ev
         else:
-            self.shard.grad.add_(out)
+            if self.ws > 1:
+                out = torch.empty(self.shard_numel, dtype=full_grad.dtype,
+                                  device=full_grad.device)
+                dist.reduce_scatter_tensor(out, full_grad, group=self.group)
+            else:
+                out = full_grad
+            if dp_group is not None:
+                dist.all_reduce(out, group=dp_group)
+            if grad_scale != 1.0:
+                out.mul_(grad_scale)
+            if self.shard.grad is None:
+                self.shard.grad = out
+            else:
+                self.shard.grad.add_(out)
         self.full_flat.grad = None
 
 
@@ -295,6 +306,7 @@ class FullyShardedDataParallel(ParallelModule):
         self._exec_order: List[FlatParamUnit] = []
         self._order_final = False
         self._backward_pending = 0
+        self._sync_scheduled = False
         if self.mesh.global_rank == 0:
             tot = sum(u.total_numel for u in self.units)
             logger.info("FSDP: %d units, %.1fM params, shard dtype %s, ws=%d",
@@ -409,6 +421,14 @@ class FullyShardedDataParallel(ParallelModule):
                 visit(o)
 
     def _post_backward_unit(self, unit: FlatParamUnit, _leaf):
+        if torch.cuda.is_available() and not self._sync_scheduled:
+            # once per backward: make the compute stream wait for every
+            # unit's grad reduction at the very end of this backward, so
+            # the optimizer (compute stream) sees finished grads while the
+            # reductions themselves overlap backward
+            self._sync_scheduled = True
+            torch.autograd.Variable._execution_engine.queue_callback(
+                self._finalize_grad_sync)
         unit.reduce_grad(self.grad_scale, self.dp_group, self._rs_stream())
         unit.in_backward = False
         unit.pending_bwd = max(0, unit.pending_bwd - 1)
@@ -416,6 +436,14 @@ class FullyShardedDataParallel(ParallelModule):
         # into this storage — reshard only after the last one
         if unit.pending_bwd == 0:
             unit.reshard()
+
+    def _finalize_grad_sync(self):
+        self._sync_scheduled = False
+        cur = torch.cuda.current_stream()
+        for u in self.units:
+            if u.rs_event is not None:
+                cur.wait_event(u.rs_event)
+                u.rs_event = None
 
     # ---- nn.Module API ---------------------------------------------------
 
