@@ -116,3 +116,40 @@ def test_gc_units_reshard_after_backward():
     for _ in range(2):
         rank, bad = q.get()
         assert not bad, f"rank {rank}: units left gathered {bad}"
+
+
+def _no_sync_worker(rank, world, q):
+    import torchacc_amd as ta
+    cfg = ta.Config()
+    cfg.dist.fsdp.size = world
+    cfg.dist.fsdp.wrap_layer_cls = {"LlamaDecoderLayer"}
+    model = _make_model()
+    wrapped = ta.accelerate(model, config=cfg)
+    fsdp = wrapped.fsdp_wrapper
+    torch.manual_seed(42)
+    a = torch.randint(0, 1024, (2, 32))
+    b = torch.randint(0, 1024, (2, 32))
+    # accumulate two micro-steps under no_sync, reduce on the third
+    with fsdp.no_sync():
+        wrapped(a, labels=a).backward()
+    wrapped(b, labels=b).backward()
+    g1 = {i: u.shard.grad.clone() for i, u in enumerate(fsdp.units)}
+    for u in fsdp.units:
+        u.shard.grad = None
+    # reference: two separate synchronized backwards (grads add)
+    wrapped(a, labels=a).backward()
+    wrapped(b, labels=b).backward()
+    ok = all(
+        torch.allclose(g1[i], u.shard.grad, atol=1e-5)
+        for i, u in enumerate(fsdp.units))
+    q.put((rank, ok))
+
+
+def test_no_sync_grad_accumulation():
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    run_multiprocess(_no_sync_worker, world_size=2, args=(q,))
+    for _ in range(2):
+        rank, ok = q.get()
+        assert ok, f"rank {rank} no_sync accumulation mismatch"

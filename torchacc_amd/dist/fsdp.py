@@ -420,7 +420,26 @@ class FullyShardedDataParallel(ParallelModule):
             for o in output.values():
                 visit(o)
 
+    import contextlib
+
+    @contextlib.contextmanager
+    def no_sync(self):
+        """Skip gradient reduction inside the context (gradient
+        accumulation): grads accumulate in the full flat grad; the first
+        backward outside the context reduces the accumulated total."""
+        self._no_sync = True
+        try:
+            yield
+        finally:
+            self._no_sync = False
+
     def _post_backward_unit(self, unit: FlatParamUnit, _leaf):
+        if getattr(self, "_no_sync", False):
+            # keep full grad + storage; reduction happens on the first
+            # synchronized backward
+            unit.in_backward = False
+            unit.pending_bwd = max(0, unit.pending_bwd - 1)
+            return
         if torch.cuda.is_available() and not self._sync_scheduled:
             # once per backward: make the compute stream wait for every
             # unit's grad reduction at the very end of this backward, so
