@@ -305,12 +305,15 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
                       const int* __restrict__ k_lens) {
   constexpr int NT = D / 16;
   constexpr int NA = D / 32;
-  constexpr int KVB = 64;
+  // 128-key staged tiles: one 8-wave WG is resident anyway (VGPR-bound),
+  // so the bigger LDS footprint (96 KB at D=128) costs no occupancy and
+  // halves the barrier + staging-iteration count.
+  constexpr int KVB = 128;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  short* k_lds = reinterpret_cast<short*>(smem);     // [64][D] swz
-  short* kt_lds = k_lds + KVB * D;                   // [D][64] swz
-  short* v_lds = kt_lds + D * KVB;                   // [64][D] swz
+  short* k_lds = reinterpret_cast<short*>(smem);     // [128][D] swz
+  short* kt_lds = k_lds + KVB * D;                   // [D][128] swz
+  short* v_lds = kt_lds + D * KVB;                   // [128][D] swz
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -403,9 +406,12 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
     }
     __syncthreads();
 
-    bool wave_active = true;
-    if (CAUSAL && kv0 > q0 + 31 + shift) wave_active = false;
-    if (HAS_WINDOW && wl >= 0 && kv0 + KVB <= q0 + shift - wl)
+#pragma unroll 1
+    for (int half = 0; half < KVB / 64; ++half) {
+    const int kv0h = kv0 + half * 64;
+    bool wave_active = kv0h < kv_hi_key;
+    if (CAUSAL && kv0h > q0 + 31 + shift) wave_active = false;
+    if (HAS_WINDOW && wl >= 0 && kv0h + 64 <= q0 + shift - wl)
       wave_active = false;
 
     if (wave_active) {
@@ -420,7 +426,7 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
       for (int kb = 0; kb < 2; ++kb) {
 #pragma unroll
         for (int t = 0; t < NT; ++t) {
-          const int row = kb * 32 + col;
+          const int row = half * 64 + kb * 32 + col;
           unsigned byte = row * (D * 2) + (t * 16 + hi * 8) * 2;
           byte ^= (unsigned)((row & 7) << 4);
           bf16x8 kf = *reinterpret_cast<const bf16x8*>(
@@ -439,7 +445,7 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
       for (int kb = 0; kb < 2; ++kb) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const int key = kv0 + kb * 32 + CROW(r, hi);
+          const int key = kv0h + kb * 32 + CROW(r, hi);
           bool valid = lane_ok && (key < klimit);
           if (CAUSAL) valid &= (key <= qrow + shift);
           if (HAS_WINDOW && wl >= 0) valid &= (key >= qrow + shift - wl);
@@ -461,7 +467,7 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
           for (int a = 0; a < NA; ++a) {
             const int d = a * 32 + col;
             unsigned byte = (unsigned)d * (KVB * 2) +
-                            (kb * 32 + 16 * tp + hi * 8) * 2;
+                            (half * 64 + kb * 32 + 16 * tp + hi * 8) * 2;
             byte ^= (unsigned)(((d >> 3) & 7) << 4);
             bf16x8 ktb = *reinterpret_cast<const bf16x8*>(
                 reinterpret_cast<const char*>(kt_lds) + byte);
@@ -471,6 +477,7 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
         }
       }
     }
+    }  // half
     __syncthreads();
   }
 
@@ -521,7 +528,7 @@ static void launch_fa_bwd(const torch::Tensor& dout, const torch::Tensor& q,
   dim3 gkv((sk + 127) / 128, b * hk), bkv(256);
   const int lds_kv = (2 * 32 * D + 2 * D * 32 + 128 * D) * 2 + 2 * 32 * 4;
   dim3 gq((sq + 255) / 256, b * hq), bq(512);
-  const int lds_q = 3 * 64 * D * 2;
+  const int lds_q = 3 * 128 * D * 2;
 
 #define DISPATCH(C, W, L)                                                    \
   do {                                                                       \

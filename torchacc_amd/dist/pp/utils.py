@@ -71,6 +71,12 @@ class SplitResult:
     final_structure: Any        # how to rebuild the output from final_vids
     batch_keys: List[str]
     attrs: Dict[str, Any] = field(default_factory=dict)
+    tied_groups: List[Dict[int, str]] = field(default_factory=list)
+    # parameters shared by >1 stage (e.g. embedding tied to the LM head):
+    # each entry maps stage_id -> qualified param name inside that stage's
+    # submodule; the executor all-reduces their grads across those stages
+    # after every backward (reference keeps tied weights consistent via the
+    # compiler; eager PP must sum the per-stage grads explicitly)
     # top-level get_attr values (buffers like rope tables) consumed by
     # stage inputs — the reference moved these into consumer submodules
     # (move_single_param_to_callee, trace.py:95-176); we hand them to the
@@ -287,11 +293,9 @@ def split(gm: fx.GraphModule, num_stages: int,
             orig = attrs[val]
             san = "_attr_" + val.replace(".", "_")
             if isinstance(orig, torch.nn.Parameter):
-                if len(attr_consumers[val]) > 1:
-                    raise NotImplementedError(
-                        f"parameter '{val}' is consumed by multiple "
-                        "pipeline stages (tied weights across stages are "
-                        "not supported yet)")
+                # shared across stages is fine: the SAME Parameter object
+                # is registered in every consuming stage and the executor
+                # sums its grad across those stages after backward
                 sub.register_parameter(san, orig)
             elif isinstance(orig, torch.Tensor):
                 sub.register_buffer(
@@ -311,10 +315,23 @@ def split(gm: fx.GraphModule, num_stages: int,
         else:
             spec.inputs = new_inputs
 
+    # ---- tied-parameter detection (same Parameter object in >1 stage,
+    # e.g. input embedding reused as the LM head weight) ----
+    by_id: Dict[int, Dict[int, str]] = {}
+    for st, sub in enumerate(submods):
+        for name, prm in sub.named_parameters():
+            by_id.setdefault(id(prm), {})[st] = name
+    tied_groups = [g for g in by_id.values() if len(g) > 1]
+    if tied_groups:
+        logger.info("pp: %d tied parameter group(s) across stages: %s",
+                    len(tied_groups),
+                    [sorted(g.keys()) for g in tied_groups])
+
     return SplitResult(submodules=submods, specs=specs,
                        final_vids=final_vids,
                        final_structure=final_structure,
-                       batch_keys=batch_keys, attrs=attrs)
+                       batch_keys=batch_keys, attrs=attrs,
+                       tied_groups=tied_groups)
 
 
 def output_index_map(sub: fx.GraphModule) -> int:
