@@ -133,3 +133,72 @@ def test_pp2_skip_connection():
         opt.zero_grad()
         base.append(float((l1 + l2) / 2))
     assert results[0] == pytest.approx(base, abs=1e-5)
+
+
+class TiedLM(torch.nn.Module):
+    """Input embedding reused as the LM-head weight: the tied parameter is
+    consumed by stage 0 (embedding) and the last stage (logits), so PP must
+    sum its grad across both stages."""
+
+    def __init__(self, vocab=64, hidden=32):
+        super().__init__()
+        self.embed = torch.nn.Embedding(vocab, hidden)
+        self.block1 = torch.nn.Linear(hidden, hidden)
+        self.block2 = torch.nn.Linear(hidden, hidden)
+
+    def forward(self, input_ids, labels=None):
+        x = torch.relu(self.block1(self.embed(input_ids)))
+        x = torch.relu(self.block2(x))
+        logits = torch.nn.functional.linear(x, self.embed.weight)
+        if labels is None:
+            return logits
+        return torch.nn.functional.cross_entropy(
+            logits.view(-1, logits.size(-1)), labels.reshape(-1))
+
+
+def _pp_tied_worker(rank, world, q):
+    import torchacc_amd as ta
+    cfg = ta.Config()
+    cfg.dist.pp.size = world
+    cfg.dist.pp.num_micro_batches = 2
+    cfg.dist.pp.input_names = ["input_ids", "labels"]
+    cfg.dist.pp.split_points = ["block2"]
+    torch.manual_seed(0)
+    model = TiedLM()
+    model = ta.accelerate(model, config=cfg)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    torch.manual_seed(7)
+    ids = torch.randint(0, 64, (4, 8))
+    losses = []
+    for _ in range(4):
+        loss = model.forward_backward(ids, labels=ids)
+        opt.step()
+        opt.zero_grad()
+        losses.append(float(loss))
+    q.put((rank, losses))
+
+
+def test_pp2_tied_embedding_matches_single_process():
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    run_multiprocess(_pp_tied_worker, world_size=2, args=(q,))
+    results = {}
+    for _ in range(2):
+        r, losses = q.get()
+        results[r] = losses
+    assert results[0] == pytest.approx(results[1], abs=1e-6)
+
+    torch.manual_seed(0)
+    model = TiedLM()
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    torch.manual_seed(7)
+    ids = torch.randint(0, 64, (4, 8))
+    base = []
+    for _ in range(4):
+        loss = model(ids, labels=ids)
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        base.append(float(loss))
+    assert results[0] == pytest.approx(base, abs=1e-5)

@@ -48,6 +48,21 @@ class PipeExecutor:
         for s, spec in enumerate(split_result.specs):
             for _idx, vid in spec.outputs:
                 self.producer_stage[vid] = s
+        # tied parameters (same Parameter object in >1 stage): sum grads
+        # across the owning stages after every backward so each stage's
+        # optimizer applies an identical update
+        self.tied_groups = []
+        for grp in split_result.tied_groups:
+            if stage_id in grp:
+                self.tied_groups.append({
+                    "name": grp[stage_id],
+                    "peers": sorted(s for s in grp if s != stage_id),
+                })
+        if self.tied_groups and config.dist.fsdp.size > 1:
+            raise NotImplementedError(
+                "tied parameters across pipeline stages are not supported "
+                "in combination with FSDP (the tied grad lives inside a "
+                "flattened shard); untie the weights or disable FSDP")
         self._reset()
 
     def _reset(self):
@@ -169,7 +184,39 @@ class PipeExecutor:
                                  self.stage_id)
         self._float_send_cache = {}
         self._exec_schedule_train(sched, output_fn)
+        self._sync_tied_grads()
         return self._aggregate_total_loss()
+
+    def _sync_tied_grads(self):
+        """Sum the gradient of every tied parameter over its owning stages
+        (p2p exchange on the pp group; contributions added in stage order so
+        all owners compute bit-identical totals)."""
+        if not self.tied_groups:
+            return
+        params = dict(self.module.named_parameters())
+        for grp in self.tied_groups:
+            p = params[grp["name"]]
+            g = p.grad if p.grad is not None else torch.zeros_like(p)
+            g = g.contiguous()
+            contrib = {self.stage_id: g}
+            ops = []
+            for s in grp["peers"]:
+                peer = self.mesh.stage_to_global(s)
+                buf = torch.empty_like(g)
+                contrib[s] = buf
+                if self.stage_id < s:
+                    ops.append(dist.P2POp(dist.isend, g, peer))
+                    ops.append(dist.P2POp(dist.irecv, buf, peer))
+                else:
+                    ops.append(dist.P2POp(dist.irecv, buf, peer))
+                    ops.append(dist.P2POp(dist.isend, g, peer))
+            for w in dist.batch_isend_irecv(ops):
+                w.wait()
+            total = None
+            for s in sorted(contrib):
+                total = contrib[s].clone() if total is None \
+                    else total + contrib[s]
+            p.grad = total
 
     def _exec_schedule_train(self, sched, output_fn):
         pending_grads: Dict[int, List[torch.Tensor]] = {}
