@@ -21,6 +21,7 @@ sources = [
         "adamw.hip",
         "flash_attn_fwd.hip",
         "flash_attn_bwd.hip",
+        "flash_attn_varlen.hip",
         "flash_attn_extra_fwd.hip",
         "flash_attn_extra_bwd.hip",
     )

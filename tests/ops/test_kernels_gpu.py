@@ -384,3 +384,68 @@ def test_fa_varlen_position_ids():
                                              causal=True)
     err = (out.float().cpu() - ref.float()).abs().max()
     assert err < 2.5e-2, err
+
+
+def test_fa_varlen_fused_matches_per_seq_loop():
+    """Fused single-launch varlen kernels vs the per-sequence loop over the
+    fixed-length kernels (same math, different tiling/masking)."""
+    from torchacc_amd.ops._backend import require_extension
+    from torchacc_amd.ops.flash_attn import _cu_to_bounds
+    ext = require_extension()
+    torch.manual_seed(0)
+    lens = [13, 256, 100, 1, 330, 64]
+    total = sum(lens)
+    cu = torch.tensor([0] + list(torch.cumsum(torch.tensor(lens), 0)),
+                      dtype=torch.int32, device="cuda")
+    for h, hk in ((4, 4), (8, 2)):
+        d = 128
+        q = torch.randn(total, h, d, device="cuda", dtype=torch.bfloat16)
+        k = torch.randn(total, hk, d, device="cuda", dtype=torch.bfloat16)
+        v = torch.randn(total, hk, d, device="cuda", dtype=torch.bfloat16)
+        do = torch.randn_like(q)
+        scale = d ** -0.5
+        bounds = _cu_to_bounds(cu, total)
+        for causal in (True, False):
+            # reference: per-sequence loop over the fixed-length kernels
+            ref_o = torch.zeros_like(q)
+            ref_lse = torch.zeros(h, total, dtype=torch.float32,
+                                  device="cuda")
+            e = torch.empty(0, device="cuda")
+            for i in range(cu.numel() - 1):
+                qs, qe = int(cu[i]), int(cu[i + 1])
+                o_i, lse_i = ext.fa_forward(
+                    q[qs:qe].unsqueeze(0).contiguous(),
+                    k[qs:qe].unsqueeze(0).contiguous(),
+                    v[qs:qe].unsqueeze(0).contiguous(), scale, causal,
+                    -1, -1, e, e, e, 0.0, 0)
+                ref_o[qs:qe] = o_i.squeeze(0)
+                ref_lse[:, qs:qe] = lse_i.squeeze(0)
+            out, lse = ext.fa_varlen_forward(q, k, v, bounds, scale, causal)
+            err = (out.float() - ref_o.float()).abs().max().item()
+            assert err < 1e-2, f"varlen fused fwd err {err} causal={causal}"
+            lerr = (lse - ref_lse).abs().max().item()
+            assert lerr < 1e-3, f"varlen fused lse err {lerr}"
+
+            ref_dq = torch.zeros_like(q)
+            ref_dk = torch.zeros_like(k)
+            ref_dv = torch.zeros_like(v)
+            for i in range(cu.numel() - 1):
+                qs, qe = int(cu[i]), int(cu[i + 1])
+                dq_i, dk_i, dv_i = ext.fa_backward(
+                    do[qs:qe].unsqueeze(0).contiguous(),
+                    q[qs:qe].unsqueeze(0).contiguous(),
+                    k[qs:qe].unsqueeze(0).contiguous(),
+                    v[qs:qe].unsqueeze(0).contiguous(),
+                    ref_o[qs:qe].unsqueeze(0).contiguous(),
+                    ref_lse[:, qs:qe].unsqueeze(0).contiguous(), scale,
+                    causal, -1, -1, e, e, e, 0.0, 0)
+                ref_dq[qs:qe] = dq_i.squeeze(0)
+                ref_dk[qs:qe] = dk_i.squeeze(0)
+                ref_dv[qs:qe] = dv_i.squeeze(0)
+            dq, dk, dv = ext.fa_varlen_backward(do, q, k, v, out, lse,
+                                                bounds, scale, causal)
+            for name, a, b in (("dq", dq, ref_dq), ("dk", dk, ref_dk),
+                               ("dv", dv, ref_dv)):
+                errg = (a.float() - b.float()).abs().max().item()
+                assert errg < 2e-2, \
+                    f"varlen fused {name} err {errg} causal={causal}"

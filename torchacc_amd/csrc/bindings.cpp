@@ -54,6 +54,18 @@ std::vector<torch::Tensor> fa_backward(torch::Tensor dout, torch::Tensor q,
                                        torch::Tensor alibi_slopes,
                                        double p_drop, long rng_seed);
 
+// flash_attn_varlen.hip
+std::vector<torch::Tensor> fa_varlen_forward(torch::Tensor q,
+                                             torch::Tensor k,
+                                             torch::Tensor v,
+                                             torch::Tensor bounds,
+                                             double softmax_scale,
+                                             bool causal);
+std::vector<torch::Tensor> fa_varlen_backward(
+    torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor out, torch::Tensor lse, torch::Tensor bounds,
+    double softmax_scale, bool causal);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "torchacc_amd CDNA4 (gfx950) kernels";
   m.def("rmsnorm_forward", &rmsnorm_forward);
@@ -68,4 +80,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adamw", &fused_adamw);
   m.def("fa_forward", &fa_forward);
   m.def("fa_backward", &fa_backward);
+  m.def("fa_varlen_forward", &fa_varlen_forward);
+  m.def("fa_varlen_backward", &fa_varlen_backward);
 }

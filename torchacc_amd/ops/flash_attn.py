@@ -324,12 +324,39 @@ class FlashAttnVarlenFunc(torch.autograd.Function):
         return (dq, dk, dv) + (None,) * 9
 
 
+def _cu_to_bounds(cu: torch.Tensor, total: int) -> torch.Tensor:
+    """cu_seqlens -> per-row int32 [total, 2] sequence intervals (the form
+    the fused varlen kernels consume: same-sequence masking collapses to a
+    per-q-row interval test, see csrc/flash_attn_varlen.hip)."""
+    cu = cu.to(torch.int32)
+    lens = (cu[1:] - cu[:-1]).long()
+    starts = torch.repeat_interleave(cu[:-1], lens)
+    ends = torch.repeat_interleave(cu[1:], lens)
+    b = torch.stack([starts, ends], dim=1)
+    if b.shape[0] < total:
+        # rows past cu[-1] (padding): empty interval -> zero out/grads
+        b = torch.cat([b, torch.zeros(total - b.shape[0], 2,
+                                      dtype=torch.int32,
+                                      device=b.device)])
+    return b.contiguous()
+
+
+def _use_fused_varlen(q, k, cu_q, cu_k, wl, wr):
+    return (wl < 0 and wr < 0 and q.shape[0] == k.shape[0] and
+            cu_q.numel() == cu_k.numel() and
+            bool(torch.equal(cu_q, cu_k)))
+
+
 def _varlen_fwd_gpu(ext, q, k, v, cu_q, cu_k, softmax_scale, causal, wl,
                     wr):
-    """Packed varlen on the fixed-length CDNA4 kernel: one launch per
-    sequence (the kernels are whole-tile; a fused multi-sequence varlen
-    kernel is a planned optimization)."""
+    """Packed varlen: ONE fused kernel launch over the whole packed batch
+    when q/k share the packing and there is no sliding window; otherwise
+    one launch per sequence on the fixed-length kernel."""
     total, h, _d = q.shape
+    if _use_fused_varlen(q, k, cu_q, cu_k, wl, wr):
+        bounds = _cu_to_bounds(cu_q.to(q.device), total)
+        return tuple(ext.fa_varlen_forward(q, k, v, bounds, softmax_scale,
+                                           causal))
     out = torch.zeros_like(q)
     lse = torch.zeros(h, total, dtype=torch.float32, device=q.device)
     empty = torch.empty(0, device=q.device)
@@ -350,6 +377,10 @@ def _varlen_fwd_gpu(ext, q, k, v, cu_q, cu_k, softmax_scale, causal, wl,
 
 def _varlen_bwd_gpu(ext, dout, q, k, v, out, lse, cu_q, cu_k, softmax_scale,
                     causal, wl, wr):
+    if _use_fused_varlen(q, k, cu_q, cu_k, wl, wr):
+        bounds = _cu_to_bounds(cu_q.to(q.device), q.shape[0])
+        return tuple(ext.fa_varlen_backward(dout, q, k, v, out, lse.contiguous(),
+                                            bounds, softmax_scale, causal))
     dq = torch.zeros_like(q)
     dk = torch.zeros_like(k)
     dv = torch.zeros_like(v)
