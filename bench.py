@@ -36,6 +36,9 @@ def parse_args():
     p.add_argument("--mode", default="fsdp",
                    choices=["fsdp", "ulysses", "ring", "2d", "fsdp_tp"])
     p.add_argument("--tp", type=int, default=1)
+    p.add_argument("--no-gc-selective", action="store_true",
+                   help="full recompute inside checkpointed layers instead "
+                        "of retaining attention outputs (selective AC)")
     p.add_argument("--no-gc", action="store_true",
                    help="disable gradient checkpointing")
     return p.parse_args()
@@ -89,6 +92,7 @@ def main():
     if not args.no_gc and args.model != "tiny":
         cfg.memory.gc = True
         cfg.memory.gc_cls = {"LlamaDecoderLayer"}
+        cfg.memory.gc_selective_attn = not args.no_gc_selective
 
     if on_gpu:
         torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", 0)))
@@ -101,6 +105,8 @@ def main():
 
     device = ta.lazy_device()
     seq = args.seq_len
+    if args.model == "tiny":
+        seq = min(seq, 512)  # tiny config's rope table / max_position
     bs = args.batch_size
     is_cp = args.mode in ("ulysses", "ring", "2d")
     local_seq = seq // world if is_cp else seq
@@ -174,7 +180,10 @@ def main():
                 "global_batch": bs * (1 if is_cp else world),
                 "seq_len": seq,
                 "parallelism": parallelism,
-                "grad_checkpoint": cfg.memory.gc,
+                "grad_checkpoint": ("selective_attn"
+                                    if cfg.memory.gc and
+                                    cfg.memory.gc_selective_attn
+                                    else cfg.memory.gc),
                 "loss": float(last) if last is not None else None,
             },
         }

@@ -75,11 +75,13 @@ def accelerate(model: torch.nn.Module, dataloader=None,
             model, ParallelModule) else model
         if config.memory.gc_cls:
             gradient_checkpoint(inner, config.memory.gc_cls,
-                                config.memory.gc_cnt)
+                                config.memory.gc_cnt,
+                                config.memory.gc_selective_attn)
         else:
             logger.warning("memory.gc set without gc_cls: wrapping the root "
                            "module in one checkpoint region")
-            wrapped = gradient_checkpoint(inner)
+            wrapped = gradient_checkpoint(
+                inner, selective_attn=config.memory.gc_selective_attn)
             if isinstance(model, ParallelModule):
                 model._update_underlay_model(wrapped)
             else:

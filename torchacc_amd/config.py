@@ -45,13 +45,19 @@ class MemoryConfig:
     gc: enable gradient checkpointing.
     gc_cls: set of module class *names* to checkpoint (empty -> root module).
     gc_cnt: checkpoint only the first ``gc_cnt`` instances (None -> all).
+    gc_selective_attn: Megatron-style selective activation checkpointing —
+        retain attention outputs (out, lse) so recomputation skips the
+        attention kernels (memory cost ~b*s*h*d bf16 per layer; the right
+        trade on 288 GB HBM3E).
     """
     gc: bool = False
     gc_cls: Set[str] = field(default_factory=set)
     gc_cnt: Optional[int] = None
+    gc_selective_attn: bool = False
 
     def validate(self):
         assert isinstance(self.gc, bool)
+        assert isinstance(self.gc_selective_attn, bool)
         if self.gc_cnt is not None:
             assert self.gc_cnt >= 0
 

@@ -19,11 +19,56 @@ context-parallel LSE merge math is unchanged. fp16/bf16 only. Dropout is
 accepted but only p=0.0 is supported by the CDNA4 kernels so far.
 """
 import math
+import threading
+from collections import deque
 from typing import Optional, Tuple
 
 import torch
 
 from ._backend import dispatch
+
+# ---- selective activation checkpointing (Megatron-style) -------------------
+# utils/checkpoint.py passes sac_contexts as torch.utils.checkpoint's
+# context_fn: the capture context retains (out, lse, seed) of every
+# attention call inside the checkpointed region, and the recompute context
+# replays them so recomputation skips the attention kernels entirely.
+
+_sac_tls = threading.local()
+
+
+class _SacContext:
+
+    def __init__(self, mode, queue):
+        self.mode = mode
+        self.queue = queue
+
+    def __enter__(self):
+        self.prev = getattr(_sac_tls, "state", None)
+        _sac_tls.state = (self.mode, self.queue)
+
+    def __exit__(self, *exc):
+        _sac_tls.state = self.prev
+        return False
+
+
+def sac_contexts():
+    """Fresh queue per checkpoint invocation: capture/recompute pairs stay
+    matched even under PP micro-batch interleaving."""
+    q = deque()
+    return _SacContext("capture", q), _SacContext("recompute", q)
+
+
+def _sac_pop():
+    state = getattr(_sac_tls, "state", None)
+    if state is not None and state[0] == "recompute" and state[1]:
+        return state[1].popleft()
+    return None
+
+
+def _sac_push(entry):
+    state = getattr(_sac_tls, "state", None)
+    if state is not None and state[0] == "capture":
+        state[1].append(entry)
 
 
 def _check_qkv(q, k, v):
@@ -98,23 +143,28 @@ class FlashAttnFunc(torch.autograd.Function):
             ext = None  # CDNA4 attention kernels are bf16
         wl, wr = window_size
         al = alibi_slopes if alibi_slopes is not None else torch.empty(0)
-        seed = 0
-        if dropout_p > 0.0:
-            assert ext is not None, \
-                "attention dropout runs only on the CDNA4 kernels (GPU)"
-            # host-RNG seed: reproducible under torch.manual_seed; the
-            # backward kernels regenerate the same keep-mask from it
-            seed = int(torch.randint(0, 2 ** 62, (1,)).item())
-        if ext is not None:
-            out, lse = ext.fa_forward(
-                q, k, v, softmax_scale, causal, wl, wr,
-                q_lens if q_lens is not None else torch.empty(0),
-                k_lens if k_lens is not None else torch.empty(0), al,
-                dropout_p, seed)
+        cached = _sac_pop()
+        if cached is not None:
+            out, lse, seed = cached
         else:
-            out, lse = _ref_attention(q, k, v, softmax_scale, causal,
-                                      (wl, wr), q_lens, k_lens,
-                                      alibi_slopes)
+            seed = 0
+            if dropout_p > 0.0:
+                assert ext is not None, \
+                    "attention dropout runs only on the CDNA4 kernels (GPU)"
+                # host-RNG seed: reproducible under torch.manual_seed; the
+                # backward kernels regenerate the same keep-mask from it
+                seed = int(torch.randint(0, 2 ** 62, (1,)).item())
+            if ext is not None:
+                out, lse = ext.fa_forward(
+                    q, k, v, softmax_scale, causal, wl, wr,
+                    q_lens if q_lens is not None else torch.empty(0),
+                    k_lens if k_lens is not None else torch.empty(0), al,
+                    dropout_p, seed)
+            else:
+                out, lse = _ref_attention(q, k, v, softmax_scale, causal,
+                                          (wl, wr), q_lens, k_lens,
+                                          alibi_slopes)
+        _sac_push((out.detach(), lse.detach(), seed))
         ctx.save_for_backward(
             q, k, v, out, lse,
             q_lens if q_lens is not None else torch.empty(0),
@@ -297,12 +347,16 @@ class FlashAttnVarlenFunc(torch.autograd.Function):
         q, k, v = [t.contiguous() for t in (q, k, v)]
         ext = dispatch(q)
         wl, wr = window_size
-        if ext is not None:
+        cached = _sac_pop()
+        if cached is not None:
+            out, lse = cached[0], cached[1]
+        elif ext is not None:
             out, lse = _varlen_fwd_gpu(ext, q, k, v, cu_q, cu_k,
                                        softmax_scale, causal, wl, wr)
         else:
             out, lse = _ref_varlen(q, k, v, cu_q, cu_k, softmax_scale,
                                    causal, (wl, wr))
+        _sac_push((out.detach(), lse.detach(), 0))
         ctx.save_for_backward(q, k, v, out, lse, cu_q, cu_k)
         ctx.meta = (max_q, max_k, softmax_scale, causal, window_size)
         return out, lse

@@ -15,12 +15,23 @@ from .logger import logger
 
 class CheckpointWrapper(torch.nn.Module):
 
-    def __init__(self, mod: torch.nn.Module):
+    def __init__(self, mod: torch.nn.Module, selective_attn: bool = False):
         super().__init__()
         self._checkpoint_wrapped_module = mod
+        self._selective_attn = selective_attn
 
     def forward(self, *args, **kwargs):
         if torch.is_grad_enabled():
+            if self._selective_attn:
+                # Megatron-style selective AC: attention outputs (out, lse)
+                # are retained during the first pass and replayed during
+                # recomputation (ops/flash_attn.py sac_contexts) — the
+                # recompute skips the attention kernels; 288 GB HBM3E makes
+                # the retained outputs cheap relative to the saved compute
+                from ..ops.flash_attn import sac_contexts
+                return checkpoint(self._checkpoint_wrapped_module, *args,
+                                  use_reentrant=False,
+                                  context_fn=sac_contexts, **kwargs)
             return checkpoint(self._checkpoint_wrapped_module, *args,
                               use_reentrant=False, **kwargs)
         return self._checkpoint_wrapped_module(*args, **kwargs)
@@ -34,10 +45,11 @@ class CheckpointWrapper(torch.nn.Module):
 
 def gradient_checkpoint(model: torch.nn.Module,
                         gc_cls: Optional[Set[str]] = None,
-                        gc_cnt: Optional[int] = None) -> torch.nn.Module:
+                        gc_cnt: Optional[int] = None,
+                        selective_attn: bool = False) -> torch.nn.Module:
     gc_cls = set(gc_cls or ())
     if not gc_cls:
-        return CheckpointWrapper(model)
+        return CheckpointWrapper(model, selective_attn)
     targets = []
     for mod in model.modules():
         if isinstance(mod, CheckpointWrapper):
@@ -50,7 +62,7 @@ def gradient_checkpoint(model: torch.nn.Module,
     for mod, name, child in targets:
         if gc_cnt is not None and count >= gc_cnt:
             break
-        setattr(mod, name, CheckpointWrapper(child))
+        setattr(mod, name, CheckpointWrapper(child, selective_attn))
         count += 1
     logger.info("gradient checkpointing: wrapped %d modules (%s)", count,
                 sorted(gc_cls))
