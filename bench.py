@@ -29,7 +29,8 @@ def parse_args():
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", default="llama-2-7b",
-                   choices=["llama-2-7b", "llama-2-70b", "tiny"])
+                   choices=["llama-2-7b", "llama-2-70b", "llama-3-8b",
+                            "tiny"])
     p.add_argument("--seq-len", type=int, default=4096)
     p.add_argument("--batch-size", type=int, default=8,
                    help="micro batch per GPU")
@@ -46,7 +47,8 @@ def parse_args():
 
 def build_model(args, cfg):
     from torchacc_amd.models import (LlamaConfig, LlamaForCausalLM,
-                                     llama_2_70b, llama_2_7b, llama_tiny)
+                                     llama_2_70b, llama_2_7b, llama_3_8b,
+                                     llama_tiny)
     cp_mode = args.mode if args.mode in ("ulysses", "ring", "2d") else None
     kw = dict(max_position_embeddings=max(args.seq_len, 4096),
               cp_mode=cp_mode)
@@ -54,6 +56,8 @@ def build_model(args, cfg):
         mcfg = llama_2_7b(**kw)
     elif args.model == "llama-2-70b":
         mcfg = llama_2_70b(**kw)
+    elif args.model == "llama-3-8b":
+        mcfg = llama_3_8b(**kw)
     else:
         mcfg = llama_tiny(cp_mode=cp_mode)
     torch.manual_seed(1234)

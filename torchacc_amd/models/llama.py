@@ -50,6 +50,14 @@ def llama_2_70b(**kw) -> "LlamaConfig":
         num_attention_heads=64, num_key_value_heads=8, **kw)
 
 
+def llama_3_8b(**kw) -> "LlamaConfig":
+    """Llama-3 8B: GQA (8 kv heads), 128k vocab, rope theta 500k."""
+    return LlamaConfig(
+        vocab_size=128256, hidden_size=4096, intermediate_size=14336,
+        num_hidden_layers=32, num_attention_heads=32, num_key_value_heads=8,
+        max_position_embeddings=8192, rope_theta=500000.0, **kw)
+
+
 def llama_tiny(**kw) -> "LlamaConfig":
     """4-layer toy config (driver config #1: tiny DP=1 CPU plumbing)."""
     return LlamaConfig(
