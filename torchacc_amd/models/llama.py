@@ -45,25 +45,31 @@ def llama_2_7b(**kw) -> "LlamaConfig":
 
 
 def llama_2_70b(**kw) -> "LlamaConfig":
-    return LlamaConfig(
+    base = dict(
         hidden_size=8192, intermediate_size=28672, num_hidden_layers=80,
-        num_attention_heads=64, num_key_value_heads=8, **kw)
+        num_attention_heads=64, num_key_value_heads=8)
+    base.update(kw)
+    return LlamaConfig(**base)
 
 
 def llama_3_8b(**kw) -> "LlamaConfig":
     """Llama-3 8B: GQA (8 kv heads), 128k vocab, rope theta 500k."""
-    return LlamaConfig(
+    base = dict(
         vocab_size=128256, hidden_size=4096, intermediate_size=14336,
         num_hidden_layers=32, num_attention_heads=32, num_key_value_heads=8,
-        max_position_embeddings=8192, rope_theta=500000.0, **kw)
+        max_position_embeddings=8192, rope_theta=500000.0)
+    base.update(kw)
+    return LlamaConfig(**base)
 
 
 def llama_tiny(**kw) -> "LlamaConfig":
     """4-layer toy config (driver config #1: tiny DP=1 CPU plumbing)."""
-    return LlamaConfig(
+    base = dict(
         vocab_size=1024, hidden_size=256, intermediate_size=688,
         num_hidden_layers=4, num_attention_heads=8, num_key_value_heads=8,
-        max_position_embeddings=512, **kw)
+        max_position_embeddings=512)
+    base.update(kw)
+    return LlamaConfig(**base)
 
 
 class LlamaAttention(nn.Module):
