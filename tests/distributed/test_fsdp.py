@@ -94,6 +94,7 @@ def _gc_reshard_worker(rank, world, q):
     cfg.dist.fsdp.wrap_layer_cls = {"LlamaDecoderLayer"}
     cfg.memory.gc = True
     cfg.memory.gc_cls = {"LlamaDecoderLayer"}
+    cfg.memory.gc_selective_attn = True  # SAC under FSDP regather
     model = _make_model()
     wrapped = ta.accelerate(model, config=cfg)
     fsdp = wrapped.fsdp_wrapper
