@@ -30,7 +30,7 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", default="llama-2-7b",
                    choices=["llama-2-7b", "llama-2-70b", "llama-3-8b",
-                            "tiny"])
+                            "qwen2-7b", "tiny"])
     p.add_argument("--seq-len", type=int, default=4096)
     p.add_argument("--batch-size", type=int, default=8,
                    help="micro batch per GPU")
@@ -58,6 +58,13 @@ def build_model(args, cfg):
         mcfg = llama_2_70b(**kw)
     elif args.model == "llama-3-8b":
         mcfg = llama_3_8b(**kw)
+    elif args.model == "qwen2-7b":
+        from torchacc_amd.models import Qwen2ForCausalLM, qwen2_7b
+        assert cp_mode is None, "qwen2 bench supports fsdp modes"
+        mcfg = qwen2_7b(
+            max_position_embeddings=max(args.seq_len, 4096))
+        torch.manual_seed(1234)
+        return Qwen2ForCausalLM(mcfg), mcfg
     else:
         mcfg = llama_tiny(cp_mode=cp_mode)
     torch.manual_seed(1234)
@@ -77,7 +84,7 @@ def main():
     parallelism = f"fsdp{world}"
     if args.mode == "fsdp":
         cfg.dist.fsdp.size = world
-        cfg.dist.fsdp.wrap_layer_cls = {"LlamaDecoderLayer"}
+        cfg.dist.fsdp.wrap_layer_cls = {"LlamaDecoderLayer", "Qwen2DecoderLayer"}
     elif args.mode in ("ulysses", "ring", "2d"):
         cfg.dist.sp.size = world
         if args.mode == "ulysses":
@@ -91,11 +98,11 @@ def main():
     elif args.mode == "fsdp_tp":
         cfg.dist.tp.size = args.tp
         cfg.dist.fsdp.size = world // args.tp
-        cfg.dist.fsdp.wrap_layer_cls = {"LlamaDecoderLayer"}
+        cfg.dist.fsdp.wrap_layer_cls = {"LlamaDecoderLayer", "Qwen2DecoderLayer"}
         parallelism = f"fsdp{world // args.tp}xtp{args.tp}"
     if not args.no_gc and args.model != "tiny":
         cfg.memory.gc = True
-        cfg.memory.gc_cls = {"LlamaDecoderLayer"}
+        cfg.memory.gc_cls = {"LlamaDecoderLayer", "Qwen2DecoderLayer"}
         cfg.memory.gc_selective_attn = not args.no_gc_selective
 
     if on_gpu:
