@@ -67,3 +67,36 @@ def test_accelerate_wraps_hf_model_fsdp():
     out.loss.backward()
     opt.step()
     assert torch.isfinite(out.loss)
+
+
+def test_hf_flash_attention_2_construction():
+    """attn_implementation='flash_attention_2' must work WITHOUT the CUDA
+    flash_attn package (our CDNA4 kernels are the implementation), and the
+    runtime path must route through our patched entry point."""
+    from torchacc_amd.utils import patch
+    assert patch.patch_fa() is True
+    import transformers.integrations.flash_attention as fa_int
+    calls = {"n": 0}
+    orig = fa_int._flash_attention_forward
+
+    def counted(*a, **k):
+        calls["n"] += 1
+        return orig(*a, **k)
+
+    fa_int._flash_attention_forward = counted
+    try:
+        from transformers import LlamaConfig, LlamaForCausalLM
+        cfg = LlamaConfig(
+            vocab_size=64, hidden_size=64, intermediate_size=128,
+            num_hidden_layers=2, num_attention_heads=4,
+            num_key_value_heads=4,
+            attn_implementation="flash_attention_2")
+        torch.manual_seed(0)
+        m = LlamaForCausalLM(cfg)
+        ids = torch.randint(0, 64, (2, 16))
+        out = m(input_ids=ids, labels=ids)
+        out.loss.backward()
+        assert torch.isfinite(out.loss)
+    finally:
+        fa_int._flash_attention_forward = orig
+    assert calls["n"] == 2, calls

@@ -81,6 +81,56 @@ def patch_fa() -> bool:
         return out
 
     mfa._flash_attention_forward = _flash_attention_forward
+    # the integrations module binds the symbol by value at import time —
+    # rebind it there too so real model forwards hit our kernels
+    try:
+        from transformers.integrations import flash_attention as _fa_int
+        _fa_int._flash_attention_forward = _flash_attention_forward
+    except ImportError:
+        pass
+    # let users request attn_implementation="flash_attention_2" without the
+    # CUDA flash_attn package: our CDNA4 kernels ARE the implementation
+    matrix = getattr(mfa, "FLASH_ATTENTION_COMPATIBILITY_MATRIX", None)
+    if isinstance(matrix, dict) and 2 in matrix:
+        matrix[2]["general_availability_check"] = lambda *a, **k: True
+        matrix[2]["pkg_availability_check"] = lambda *a, **k: True
+    # transformers >= 5 lazily resolves flash_attn_func/flash_attn_varlen_fn
+    # globals at model init (lazy_import_flash_attention) — pre-populate
+    # them with adapters over our kernels so construction with
+    # attn_implementation="flash_attention_2" needs no flash_attn package
+    if hasattr(mfa, "_loaded_implementation"):
+        from ..ops.flash_attn import flash_attn_varlen_func as _ta_varlen
+
+        def _ta_fa_func(q, k, v, dropout_p=0.0, softmax_scale=None,
+                        causal=False, window_size=(-1, -1),
+                        alibi_slopes=None, deterministic=False,
+                        return_attn_probs=False, **kw):
+            return flash_attn_xla(q, k, v, dropout_p=dropout_p,
+                                  softmax_scale=softmax_scale, causal=causal,
+                                  window_size=window_size,
+                                  alibi_slopes=alibi_slopes)
+
+        def _ta_fa_varlen_func(q, k, v, cu_seqlens_q, cu_seqlens_k,
+                               max_seqlen_q, max_seqlen_k, dropout_p=0.0,
+                               softmax_scale=None, causal=False,
+                               window_size=(-1, -1), alibi_slopes=None,
+                               deterministic=False, return_attn_probs=False,
+                               **kw):
+            return _ta_varlen(q, k, v, cu_seqlens_q, cu_seqlens_k,
+                              max_seqlen_q, max_seqlen_k,
+                              dropout_p=dropout_p,
+                              softmax_scale=softmax_scale, causal=causal,
+                              window_size=window_size)
+
+        try:
+            mfa._loaded_implementation = "flash_attention_2"
+            mfa._flash_fn = _ta_fa_func
+            mfa._flash_varlen_fn = _ta_fa_varlen_func
+            if hasattr(mfa, "_lazy_define_process_function"):
+                mfa._process_flash_kwargs_fn = \
+                    mfa._lazy_define_process_function(_ta_fa_varlen_func)
+        except Exception:  # version drift: our _flash_attention_forward
+            pass           # replacement still covers the runtime path
     logger.info("patched transformers flash-attention entry point")
     return True
 
