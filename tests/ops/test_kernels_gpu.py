@@ -474,9 +474,9 @@ def test_hf_llama_uses_cdna4_kernels_gpu():
     ext.fa_forward = counted
     try:
         cfg = HFLlamaConfig(
-            vocab_size=512, hidden_size=256, intermediate_size=688,
-            num_hidden_layers=2, num_attention_heads=8,
-            num_key_value_heads=8, max_position_embeddings=256,
+            vocab_size=512, hidden_size=512, intermediate_size=688,
+            num_hidden_layers=2, num_attention_heads=4,
+            num_key_value_heads=4, max_position_embeddings=256,
             attn_implementation="flash_attention_2")
         torch.manual_seed(0)
         model = HFLlamaForCausalLM(cfg).cuda().to(torch.bfloat16)
