@@ -71,6 +71,30 @@ def _sac_push(entry):
         state[1].append(entry)
 
 
+_warned_dims = set()
+
+
+def _kernel_ext(q):
+    """The CDNA4 kernels cover bf16 with head_dim 64/128; anything else
+    runs the fp32 composite (warned once per dim, never silently on the
+    hot path — the flagship models are all head_dim 128)."""
+    ext = dispatch(q)
+    if ext is None:
+        return None
+    if q.dtype != torch.bfloat16:
+        return None
+    d = q.shape[-1]
+    if d not in (64, 128):
+        if d not in _warned_dims:
+            _warned_dims.add(d)
+            from ..utils.logger import logger
+            logger.warning(
+                "flash-attention head_dim %d unsupported by the CDNA4 "
+                "kernels (64/128); using the composite fallback", d)
+        return None
+    return ext
+
+
 def _check_qkv(q, k, v):
     if q.is_cuda:
         assert q.dtype in (torch.float16, torch.bfloat16, torch.float32), \
@@ -138,9 +162,7 @@ class FlashAttnFunc(torch.autograd.Function):
         if softmax_scale is None:
             softmax_scale = 1.0 / math.sqrt(q.shape[-1])
         q, k, v = [t.contiguous() for t in (q, k, v)]
-        ext = dispatch(q)
-        if q.dtype != torch.bfloat16:
-            ext = None  # CDNA4 attention kernels are bf16
+        ext = _kernel_ext(q)
         wl, wr = window_size
         al = alibi_slopes if alibi_slopes is not None else torch.empty(0)
         cached = _sac_pop()
@@ -181,9 +203,7 @@ class FlashAttnFunc(torch.autograd.Function):
         q, k, v, out, lse, q_lens, k_lens, al = ctx.saved_tensors
         q_lens = q_lens if q_lens.numel() else None
         k_lens = k_lens if k_lens.numel() else None
-        ext = dispatch(q)
-        if q.dtype != torch.bfloat16:
-            ext = None
+        ext = _kernel_ext(q)
         dout = dout.contiguous()
         wl, wr = ctx.window
         if ext is not None:
@@ -345,7 +365,7 @@ class FlashAttnVarlenFunc(torch.autograd.Function):
         if softmax_scale is None:
             softmax_scale = 1.0 / math.sqrt(q.shape[-1])
         q, k, v = [t.contiguous() for t in (q, k, v)]
-        ext = dispatch(q)
+        ext = _kernel_ext(q)
         wl, wr = window_size
         cached = _sac_pop()
         if cached is not None:
@@ -365,7 +385,7 @@ class FlashAttnVarlenFunc(torch.autograd.Function):
     def backward(ctx, dout, _dlse):
         q, k, v, out, lse, cu_q, cu_k = ctx.saved_tensors
         max_q, max_k, softmax_scale, causal, window_size = ctx.meta
-        ext = dispatch(q)
+        ext = _kernel_ext(q)
         dout = dout.contiguous()
         wl, wr = window_size
         if ext is not None:
