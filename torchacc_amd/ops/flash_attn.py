@@ -416,9 +416,12 @@ def _cu_to_bounds(cu: torch.Tensor, total: int) -> torch.Tensor:
 
 
 def _use_fused_varlen(q, k, cu_q, cu_k, wl, wr):
+    # `cu_q is cu_k` short-circuit: the position-ids and qkv-packed paths
+    # pass the same tensor for both, avoiding a device sync (torch.equal)
+    # on every attention call
     return (wl < 0 and wr < 0 and q.shape[0] == k.shape[0] and
             cu_q.numel() == cu_k.numel() and
-            bool(torch.equal(cu_q, cu_k)))
+            (cu_q is cu_k or bool(torch.equal(cu_q, cu_k))))
 
 
 def _varlen_fwd_gpu(ext, q, k, v, cu_q, cu_k, softmax_scale, causal, wl,
