@@ -23,6 +23,7 @@ from . import ops
 from . import utils
 from . import llm
 from .hf_trainer import accelerate_hf_trainer
+from .llm.qwen_patch import patch_qwen_model  # noqa: F401
 from .utils.logger import logger
 
 __version__ = "0.1.0"
