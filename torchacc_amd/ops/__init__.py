@@ -14,3 +14,10 @@ from .scaled_dot_product_attention import scaled_dot_product_attention  # noqa: 
 from . import context_parallel  # noqa: F401
 from .context_parallel import (  # noqa: F401
     context_parallel_2d, ring_attention, ulysses)
+
+# reference-compatible fused-kernel entry points (reference ops/liger.py;
+# ours are HIP kernels, not Triton)
+from ..utils.patch import (apply_liger_kernel,  # noqa: F401,E402
+                           apply_liger_kernel_to_llama,
+                           apply_liger_kernel_to_qwen2,
+                           apply_fused_kernel_patches)

@@ -176,8 +176,10 @@ def apply_fused_kernel_patches(model: Optional[torch.nn.Module] = None
     return patched
 
 
-# reference-compatible alias
+# reference-compatible aliases (ops/liger.py:133 exposed per-model names)
 apply_liger_kernel = apply_fused_kernel_patches
+apply_liger_kernel_to_llama = apply_fused_kernel_patches
+apply_liger_kernel_to_qwen2 = apply_fused_kernel_patches
 
 
 def patch_llama(use_flash_attn: bool = True) -> bool:
