@@ -110,6 +110,7 @@ def rendezvous(tag: str = "") -> None:
 
 
 # strategy wrappers (imported late to avoid cycles)
+from .parallel_module import ParallelModule  # noqa: E402,F401
 from .dp import DataParallel  # noqa: E402
 from .fsdp import FullyShardedDataParallel  # noqa: E402
 from .spmd_fsdp import SpmdFullyShardedDataParallel  # noqa: E402
@@ -117,3 +118,9 @@ from .distributed_parallel import DistributedParallel  # noqa: E402
 from .pp.pipeline import PipelineParallel  # noqa: E402
 from . import tp  # noqa: E402
 from . import state_dict_utils  # noqa: E402
+
+# reference-compatible backend-name constants: on the eager MI355X build
+# both map to RCCL ("nccl" IS RCCL on ROCm); the reference distinguished
+# its registered 'lazy' c10d backend from eager NCCL (dist/__init__.py:28-29)
+BACKEND_NAME = "nccl"
+EAGER_BACKEND_NAME = "nccl"
