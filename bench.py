@@ -196,6 +196,9 @@ def main():
                                     cfg.memory.gc_selective_attn
                                     else cfg.memory.gc),
                 "loss": float(last) if last is not None else None,
+                "peak_mem_gb": (round(
+                    torch.cuda.max_memory_allocated() / 2**30, 2)
+                    if on_gpu else None),
             },
         }
         print(json.dumps(out))
