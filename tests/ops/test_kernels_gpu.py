@@ -495,3 +495,39 @@ def test_hf_llama_uses_cdna4_kernels_gpu():
     assert calls["n"] >= 10, \
         f"flash-attention extension not used by HF path ({calls['n']})"
     assert losses[-1] < losses[0], losses
+
+
+def test_cpu_offload_async_matches_baseline_gpu():
+    """Async double-buffered activation offload (d2h/h2d side streams) must
+    not change gradients."""
+    from torchacc_amd.utils.cpu_offload import get_cpu_offload_context
+
+    def run(offload):
+        torch.manual_seed(0)
+        layers = torch.nn.ModuleList(
+            [torch.nn.Linear(256, 256) for _ in range(6)]).cuda()
+        torch.manual_seed(1)
+        x0 = torch.randn(4, 128, 256, device="cuda")
+        if offload:
+            ctx, commit = get_cpu_offload_context(
+                num_offload_layers=4, num_prefetch_layers=1)
+            x = x0
+            for layer in layers:
+                with ctx:
+                    x = torch.relu(layer(x))
+                x = commit(x)
+        else:
+            x = x0
+            for layer in layers:
+                x = torch.relu(layer(x))
+        loss = x.square().mean()
+        loss.backward()
+        torch.cuda.synchronize()
+        return float(loss), [p.grad.clone() for p in layers.parameters()]
+
+    l_ref, g_ref = run(False)
+    l_off, g_off = run(True)
+    assert l_ref == l_off
+    for a, b in zip(g_ref, g_off):
+        assert torch.allclose(a, b, atol=1e-6), \
+            (a - b).abs().max().item()
