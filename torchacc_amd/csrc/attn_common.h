@@ -34,6 +34,30 @@ DEVINLINE void t12_pack_frag(const V16& p16 /*16 regs*/, int tp,
 }
 
 
+typedef unsigned attn_u32x2 __attribute__((ext_vector_type(2)));
+typedef unsigned attn_u32x4 __attribute__((ext_vector_type(4)));
+
+// gfx950 ds_read_b64_tr_b16: each lane fetches 4 contiguous bf16 (8 B) at
+// its own LDS byte address; the hardware transposes the 16-lane group's
+// [16 lane][4 elem] fetch matrix to [4][16], delivering column (lane&15).
+// Measured semantics (csrc/tools/tr16_probe.hip): out[m][j] = the element
+// fetched by lane ((m+16j)>>2) at slot ((m+16j)&3) of the same group —
+// so with lane addresses  addr(l) = &X[q0 + ((l&15)>>2)][d0 + (l&3)*4]
+// the read delivers frag slots j=0..3 of the B-fragment column
+// X[q0+j][d0 + (l&15)] (row-position swizzles are free: addresses are
+// per-lane). Two reads (q0, q0+4) build a full bf16x8 B-frag.
+DEVINLINE attn_u32x2 attn_tr16_pair(unsigned a0, unsigned a1,
+                                    attn_u32x2& hi) {
+  attn_u32x2 lo;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=v"(lo), "=v"(hi)
+      : "v"(a0), "v"(a1));
+  return lo;
+}
+
 // counter-based dropout RNG (splitmix64): deterministic keep-decision per
 // (seed, flat attention index) — the backward kernels regenerate the same
 // mask the forward applied (reference: FA2 philox rng_state).
