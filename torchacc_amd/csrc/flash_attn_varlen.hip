@@ -318,14 +318,11 @@ void fa_vl_dkv_kernel(const short* __restrict__ dOut,
   constexpr int NA = D / 32;
   constexpr int QT = 32;
   constexpr int KVWG = 128;
-  constexpr int TS = 32;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* q_lds = reinterpret_cast<short*>(smem);
   short* do_lds = q_lds + QT * D;
-  short* qt_lds = do_lds + QT * D;
-  short* dot_lds = qt_lds + D * TS;
-  short* v_lds = dot_lds + D * TS;
+  short* v_lds = do_lds + QT * D;
   float* lse_lds = reinterpret_cast<float*>(v_lds + KVWG * D);
   float* del_lds = lse_lds + QT;
   int2* bnd_lds = reinterpret_cast<int2*>(del_lds + QT);
@@ -376,6 +373,14 @@ void fa_vl_dkv_kernel(const short* __restrict__ dOut,
     __syncthreads();
   }
 
+  // per-lane constant address for the ds_read_b64_tr_b16 B-frag reads
+  // (same derivation as flash_attn_bwd.hip DKV_TR_STEP)
+  const int m4 = (lane & 15) >> 2;
+  const int dl0 = ((lane >> 4) & 1) * 16 + (lane & 3) * 4;
+  const unsigned tr_addr = (unsigned)(size_t)q_lds +
+      (unsigned)((hi * 8 + m4) * (D * 2)) +
+      (((unsigned)(dl0 * 2)) ^ ((unsigned)(m4 << 4)));
+
   f32x16 dkacc[NA], dvacc[NA];
 #pragma unroll
   for (int a = 0; a < NA; ++a) {
@@ -413,15 +418,6 @@ void fa_vl_dkv_kernel(const short* __restrict__ dOut,
               qv;
           *reinterpret_cast<s16x8*>(reinterpret_cast<char*>(do_lds) + byte) =
               dv8;
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            unsigned tb = (unsigned)(d0 + j) * (TS * 2) + row * 2;
-            tb ^= (unsigned)((((d0 + j) >> 3) & 3) << 4);
-            *reinterpret_cast<short*>(
-                reinterpret_cast<char*>(qt_lds) + tb) = qv[j];
-            *reinterpret_cast<short*>(
-                reinterpret_cast<char*>(dot_lds) + tb) = dv8[j];
-          }
         }
         for (int r = tid; r < QT; r += 256) {
           const int qrow = q0 + r;
@@ -471,6 +467,32 @@ void fa_vl_dkv_kernel(const short* __restrict__ dOut,
           s[r] = p;
           dp[r] = p * (dp[r] - del_q) * scale;
         }
+#define VL_TR_STEP(tp_, a_)                                                \
+        {                                                                  \
+          attn_u32x2 ql_, qh_, dl_, dh_;                                   \
+          asm volatile(                                                    \
+              "ds_read_b64_tr_b16 %0, %4 offset:%c5\n\t"                 \
+              "ds_read_b64_tr_b16 %1, %4 offset:%c6\n\t"                 \
+              "ds_read_b64_tr_b16 %2, %4 offset:%c7\n\t"                 \
+              "ds_read_b64_tr_b16 %3, %4 offset:%c8\n\t"                 \
+              "s_waitcnt lgkmcnt(0)"                                       \
+              : "=v"(ql_), "=v"(qh_), "=v"(dl_), "=v"(dh_)                 \
+              : "v"(tr_addr),                                              \
+                "i"((tp_) * 16 * D * 2 + (a_) * 64),                       \
+                "i"((tp_) * 16 * D * 2 + 4 * D * 2 + ((a_) ^ 1) * 64),     \
+                "i"(QT * D * 2 + (tp_) * 16 * D * 2 + (a_) * 64),          \
+                "i"(QT * D * 2 + (tp_) * 16 * D * 2 + 4 * D * 2 +          \
+                    ((a_) ^ 1) * 64));                                     \
+          attn_u32x4 uq_ = {ql_.x, ql_.y, qh_.x, qh_.y};                   \
+          attn_u32x4 ud_ = {dl_.x, dl_.y, dh_.x, dh_.y};                   \
+          bf16x8 qbf = __builtin_bit_cast(bf16x8, uq_);                    \
+          bf16x8 dob = __builtin_bit_cast(bf16x8, ud_);                    \
+          dvacc[a_] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(             \
+              pb, dob, dvacc[a_], 0, 0, 0);                                \
+          dkacc[a_] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(             \
+              dsb, qbf, dkacc[a_], 0, 0, 0);                               \
+        }
+
 #pragma unroll
         for (int tp = 0; tp < 2; ++tp) {
           unsigned pfr[4], dsfr[4];
@@ -478,22 +500,23 @@ void fa_vl_dkv_kernel(const short* __restrict__ dOut,
           t12_pack_frag(dp, tp, dsfr);
           bf16x8 pb = *reinterpret_cast<const bf16x8*>(pfr);
           bf16x8 dsb = *reinterpret_cast<const bf16x8*>(dsfr);
-#pragma unroll
-          for (int a = 0; a < NA; ++a) {
-            const int d = a * 32 + col;
-            const int qoff = 16 * tp + hi * 8;
-            unsigned tb = (unsigned)d * (TS * 2) + qoff * 2;
-            tb ^= (unsigned)(((d >> 3) & 3) << 4);
-            bf16x8 dob = *reinterpret_cast<const bf16x8*>(
-                reinterpret_cast<const char*>(dot_lds) + tb);
-            bf16x8 qbf = *reinterpret_cast<const bf16x8*>(
-                reinterpret_cast<const char*>(qt_lds) + tb);
-            dvacc[a] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                pb, dob, dvacc[a], 0, 0, 0);
-            dkacc[a] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                dsb, qbf, dkacc[a], 0, 0, 0);
+          if (tp == 0) {
+            VL_TR_STEP(0, 0);
+            VL_TR_STEP(0, 1);
+            if constexpr (NA > 2) {
+              VL_TR_STEP(0, 2);
+              VL_TR_STEP(0, 3);
+            }
+          } else {
+            VL_TR_STEP(1, 0);
+            VL_TR_STEP(1, 1);
+            if constexpr (NA > 2) {
+              VL_TR_STEP(1, 2);
+              VL_TR_STEP(1, 3);
+            }
           }
         }
+#undef VL_TR_STEP
       }
       __syncthreads();
     }
@@ -534,8 +557,7 @@ void fa_vl_dq_kernel(const short* __restrict__ dOut,
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* k_lds = reinterpret_cast<short*>(smem);
-  short* kt_lds = k_lds + KVB * D;
-  short* v_lds = kt_lds + D * KVB;
+  short* v_lds = k_lds + KVB * D;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -579,6 +601,14 @@ void fa_vl_dq_kernel(const short* __restrict__ dOut,
   const float del_q =
       row_ok ? DELTA[(long)h * total + qrow] : 0.f;
 
+  // per-lane constant for ds_read_b64_tr_b16 K B-frag reads (same
+  // derivation as flash_attn_bwd.hip)
+  const int m4 = (lane & 15) >> 2;
+  const int dl0 = ((lane >> 4) & 1) * 16 + (lane & 3) * 4;
+  const unsigned ktr_base = (unsigned)(size_t)k_lds +
+      (unsigned)((hi * 8 + m4) * (D * 2)) +
+      (((unsigned)(dl0 * 2)) ^ ((unsigned)(m4 << 4)));
+
   f32x16 dqacc[NA];
 #pragma unroll
   for (int a = 0; a < NA; ++a) dqacc[a] = f32x16(0.f);
@@ -610,13 +640,6 @@ void fa_vl_dq_kernel(const short* __restrict__ dOut,
             kv8;
         *reinterpret_cast<s16x8*>(reinterpret_cast<char*>(v_lds) + byte) =
             vv8;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          unsigned tbyte = (unsigned)(d0 + j) * (KVB * 2) + row * 2;
-          tbyte ^= (unsigned)((((d0 + j) >> 3) & 7) << 4);
-          *reinterpret_cast<short*>(reinterpret_cast<char*>(kt_lds) +
-                                    tbyte) = kv8[j];
-        }
       }
     }
     __syncthreads();
@@ -664,26 +687,43 @@ void fa_vl_dq_kernel(const short* __restrict__ dOut,
           dp[kb][r] = p * (dp[kb][r] - del_q) * scale;
         }
       }
-#pragma unroll
-      for (int kb = 0; kb < 2; ++kb) {
-#pragma unroll
-        for (int tp = 0; tp < 2; ++tp) {
-          unsigned dsfr[4];
-          t12_pack_frag(dp[kb], tp, dsfr);
-          bf16x8 dsb = *reinterpret_cast<const bf16x8*>(dsfr);
-#pragma unroll
-          for (int a = 0; a < NA; ++a) {
-            const int d = a * 32 + col;
-            unsigned byte = (unsigned)d * (KVB * 2) +
-                            (half * 64 + kb * 32 + 16 * tp + hi * 8) * 2;
-            byte ^= (unsigned)(((d >> 3) & 7) << 4);
-            bf16x8 ktb = *reinterpret_cast<const bf16x8*>(
-                reinterpret_cast<const char*>(kt_lds) + byte);
-            dqacc[a] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                dsb, ktb, dqacc[a], 0, 0, 0);
-          }
-        }
+      const unsigned ktr_addr = ktr_base + (unsigned)(half * 64 * D * 2);
+
+#define VLQ_TR_STEP(kb_, tp_, a_)                                          \
+      {                                                                    \
+        attn_u32x2 kl_, kh_;                                               \
+        asm volatile(                                                      \
+            "ds_read_b64_tr_b16 %0, %2 offset:%c3\n\t"                   \
+            "ds_read_b64_tr_b16 %1, %2 offset:%c4\n\t"                   \
+            "s_waitcnt lgkmcnt(0)"                                         \
+            : "=v"(kl_), "=v"(kh_)                                         \
+            : "v"(ktr_addr),                                               \
+              "i"((kb_) * 32 * D * 2 + (tp_) * 16 * D * 2 + (a_) * 64),    \
+              "i"((kb_) * 32 * D * 2 + (tp_) * 16 * D * 2 + 4 * D * 2 +    \
+                  ((a_) ^ 1) * 64));                                       \
+        attn_u32x4 uk_ = {kl_.x, kl_.y, kh_.x, kh_.y};                     \
+        bf16x8 ktb = __builtin_bit_cast(bf16x8, uk_);                      \
+        dqacc[a_] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(               \
+            dsb, ktb, dqacc[a_], 0, 0, 0);                                 \
       }
+#define VLQ_TR_TP(kb_, tp_)                                                \
+      {                                                                    \
+        unsigned dsfr[4];                                                  \
+        t12_pack_frag(dp[kb_], tp_, dsfr);                                 \
+        bf16x8 dsb = *reinterpret_cast<const bf16x8*>(dsfr);               \
+        VLQ_TR_STEP(kb_, tp_, 0);                                          \
+        VLQ_TR_STEP(kb_, tp_, 1);                                          \
+        if constexpr (NA > 2) {                                            \
+          VLQ_TR_STEP(kb_, tp_, 2);                                        \
+          VLQ_TR_STEP(kb_, tp_, 3);                                        \
+        }                                                                  \
+      }
+      VLQ_TR_TP(0, 0);
+      VLQ_TR_TP(0, 1);
+      VLQ_TR_TP(1, 0);
+      VLQ_TR_TP(1, 1);
+#undef VLQ_TR_TP
+#undef VLQ_TR_STEP
     }
     }  // half
     __syncthreads();
@@ -780,8 +820,8 @@ std::vector<torch::Tensor> fa_varlen_backward(
 #define LAUNCH_B(DD, C)                                                      \
   do {                                                                       \
     const int lds_kv =                                                       \
-        (2 * 32 * DD + 2 * DD * 32 + 128 * DD) * 2 + 2 * 32 * 4 + 32 * 8;    \
-    const int lds_q = 3 * 128 * DD * 2;                                      \
+        (2 * 32 * DD + 128 * DD) * 2 + 2 * 32 * 4 + 32 * 8;                  \
+    const int lds_q = 2 * 128 * DD * 2;                                      \
     hipLaunchKernelGGL((fa_vl_dkv_kernel<DD, C>), gkv, bkv, lds_kv, stream,  \
                        (const short*)dout.data_ptr(),                        \
                        (const short*)q.data_ptr(),                           \
