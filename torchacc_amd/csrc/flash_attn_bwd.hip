@@ -342,7 +342,8 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* k_lds = reinterpret_cast<short*>(smem);     // [128][D] swz
-  short* v_lds = k_lds + KVB * D;                    // [128][D] swz
+  short* kt_lds = k_lds + KVB * D;                   // [D][128] swz
+  short* v_lds = kt_lds + D * KVB;                   // [128][D] swz
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -391,14 +392,6 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
   const float del_q =
       row_ok ? DELTA[((long)b * hq + h) * sq + qrow] : 0.f;
 
-  // per-lane constant for ds_read_b64_tr_b16 K B-frag reads (same
-  // derivation as the dkv DKV_TR_STEP; roles q->key)
-  const int m4 = (lane & 15) >> 2;
-  const int dl0 = ((lane >> 4) & 1) * 16 + (lane & 3) * 4;
-  const unsigned ktr_base = (unsigned)(size_t)k_lds +
-      (unsigned)((hi * 8 + m4) * (D * 2)) +
-      (((unsigned)(dl0 * 2)) ^ ((unsigned)(m4 << 4)));
-
   f32x16 dqacc[NA];
 #pragma unroll
   for (int a = 0; a < NA; ++a) dqacc[a] = f32x16(0.f);
@@ -432,6 +425,13 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
             kv8;
         *reinterpret_cast<s16x8*>(reinterpret_cast<char*>(v_lds) + byte) =
             vv8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          unsigned tbyte = (unsigned)(d0 + j) * (KVB * 2) + row * 2;
+          tbyte ^= (unsigned)((((d0 + j) >> 3) & 7) << 4);
+          *reinterpret_cast<short*>(reinterpret_cast<char*>(kt_lds) +
+                                    tbyte) = kv8[j];
+        }
       }
     }
     __syncthreads();
@@ -486,43 +486,26 @@ void fa_bwd_dq_kernel(const short* __restrict__ dOut,
         }
       }
       // dQ[q][d] += dS[q][key] K[key][d]
-      const unsigned ktr_addr = ktr_base + (unsigned)(half * 64 * D * 2);
-
-#define DQ_TR_STEP(kb_, tp_, a_)                                           \
-      {                                                                    \
-        attn_u32x2 kl_, kh_;                                               \
-        asm volatile(                                                      \
-            "ds_read_b64_tr_b16 %0, %2 offset:%c3\n\t"                   \
-            "ds_read_b64_tr_b16 %1, %2 offset:%c4\n\t"                   \
-            "s_waitcnt lgkmcnt(0)"                                         \
-            : "=v"(kl_), "=v"(kh_)                                         \
-            : "v"(ktr_addr),                                               \
-              "i"((kb_) * 32 * D * 2 + (tp_) * 16 * D * 2 + (a_) * 64),    \
-              "i"((kb_) * 32 * D * 2 + (tp_) * 16 * D * 2 + 4 * D * 2 +    \
-                  ((a_) ^ 1) * 64));                                       \
-        attn_u32x4 uk_ = {kl_.x, kl_.y, kh_.x, kh_.y};                     \
-        bf16x8 ktb = __builtin_bit_cast(bf16x8, uk_);                      \
-        dqacc[a_] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(               \
-            dsb, ktb, dqacc[a_], 0, 0, 0);                                 \
+#pragma unroll
+      for (int kb = 0; kb < 2; ++kb) {
+#pragma unroll
+        for (int tp = 0; tp < 2; ++tp) {
+          unsigned dsfr[4];
+          t12_pack_frag(dp[kb], tp, dsfr);
+          bf16x8 dsb = *reinterpret_cast<const bf16x8*>(dsfr);
+#pragma unroll
+          for (int a = 0; a < NA; ++a) {
+            const int d = a * 32 + col;
+            unsigned byte = (unsigned)d * (KVB * 2) +
+                            (half * 64 + kb * 32 + 16 * tp + hi * 8) * 2;
+            byte ^= (unsigned)(((d >> 3) & 7) << 4);
+            bf16x8 ktb = *reinterpret_cast<const bf16x8*>(
+                reinterpret_cast<const char*>(kt_lds) + byte);
+            dqacc[a] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                dsb, ktb, dqacc[a], 0, 0, 0);
+          }
+        }
       }
-#define DQ_TR_TP(kb_, tp_)                                                 \
-      {                                                                    \
-        unsigned dsfr[4];                                                  \
-        t12_pack_frag(dp[kb_], tp_, dsfr);                                 \
-        bf16x8 dsb = *reinterpret_cast<const bf16x8*>(dsfr);               \
-        DQ_TR_STEP(kb_, tp_, 0);                                           \
-        DQ_TR_STEP(kb_, tp_, 1);                                           \
-        if constexpr (NA > 2) {                                            \
-          DQ_TR_STEP(kb_, tp_, 2);                                         \
-          DQ_TR_STEP(kb_, tp_, 3);                                         \
-        }                                                                  \
-      }
-      DQ_TR_TP(0, 0);
-      DQ_TR_TP(0, 1);
-      DQ_TR_TP(1, 0);
-      DQ_TR_TP(1, 1);
-#undef DQ_TR_TP
-#undef DQ_TR_STEP
     }
     }  // half
     __syncthreads();
@@ -575,7 +558,7 @@ static void launch_fa_bwd(const torch::Tensor& dout, const torch::Tensor& q,
   dim3 gkv((sk + 127) / 128, b * hk), bkv(256);
   const int lds_kv = (2 * 32 * D + 128 * D) * 2 + 2 * 32 * 4;
   dim3 gq((sq + 255) / 256, b * hq), bq(512);
-  const int lds_q = 2 * 128 * D * 2;
+  const int lds_q = 3 * 128 * D * 2;
 
 #define DISPATCH(C, W, L)                                                    \
   do {                                                                       \
