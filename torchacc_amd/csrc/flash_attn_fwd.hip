@@ -19,7 +19,7 @@
 // per-batch varlen (q_lens/k_lens), D in {64, 128}. bf16 only.
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
-#include "attn_common.h"
+#include "common.h"
 
 #define LOG2E 1.4426950408889634f
 
@@ -48,11 +48,10 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
   constexpr int KVB = 64;      // keys per tile
   constexpr int KROW_BYTES = D * 2;
 
-  // LDS: K [64][D] + V [64][D] (both natural, (row&7)<<4 swizzle; the PV
-  // A-fragments read V TRANSPOSED via ds_read_b64_tr_b16), double buffered
+  // LDS: K [64][D] + V^T [D][64], double buffered
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* k_lds = reinterpret_cast<short*>(smem);                // 2*64*D
-  short* v_lds = reinterpret_cast<short*>(smem) + 2 * KVB * D;  // 2*64*D
+  short* vt_lds = reinterpret_cast<short*>(smem) + 2 * KVB * D; // 2*D*64
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -133,30 +132,31 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
 
   auto stage_write = [&](int buf) {
     short* kdst = k_lds + buf * KVB * D;
-    short* vdst = v_lds + buf * KVB * D;
+    short* vdst = vt_lds + buf * KVB * D;
 #pragma unroll
     for (int i = 0; i < PER_THR; ++i) {
       const int c = tid + i * 512;
       const int row = c / (D / 8);
       const int d0 = (c % (D / 8)) * 8;
-      // K and V: row-major with the (row&7)<<4 byte-XOR swizzle
-      unsigned byte = row * KROW_BYTES + d0 * 2;
-      byte ^= (unsigned)((row & 7) << 4);
-      *reinterpret_cast<s16x8*>(
-          reinterpret_cast<char*>(kdst) + byte) = kreg[i];
-      *reinterpret_cast<s16x8*>(
-          reinterpret_cast<char*>(vdst) + byte) = vreg[i];
+      // K: row-major with (row&7)<<4 byte-XOR swizzle
+      {
+        unsigned byte = row * KROW_BYTES + d0 * 2;
+        byte ^= (unsigned)((row & 7) << 4);
+        *reinterpret_cast<s16x8*>(
+            reinterpret_cast<char*>(kdst) + byte) = kreg[i];
+      }
+      // V^T: [D][64] with (d&7)<<4 swizzle, scalar scatter
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        unsigned byte = (unsigned)(d0 + j) * (KVB * 2) + row * 2;
+        byte ^= (unsigned)((((d0 + j) >> 3) & 7) << 4);
+        *reinterpret_cast<short*>(
+            reinterpret_cast<char*>(vdst) + byte) = vreg[i][j];
+      }
     }
   };
 
   // ---- online softmax state ---------------------------------------------
-  // per-lane constant for the PV tr_b16 reads
-  const int m4v = (lane & 15) >> 2;
-  const int dl0v = ((lane >> 4) & 1) * 16 + (lane & 3) * 4;
-  const unsigned vtr_base = (unsigned)(size_t)v_lds +
-      (unsigned)((hi * 8 + m4v) * (D * 2)) +
-      (((unsigned)(dl0v * 2)) ^ ((unsigned)(m4v << 4)));
-
   float m_run = -INFINITY;
   float l_run = 0.f;
   f32x16_ oacc[NA];
@@ -184,6 +184,7 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
 
     if (wave_active) {
       const short* kbuf = k_lds + buf * KVB * D;
+      const short* vbuf = vt_lds + buf * KVB * D;
       // ---- QK^T: 2 key sub-blocks x NT d-steps --------------------------
       f32x16_ p[2];
       p[0] = f32x16_(0.f);
@@ -274,41 +275,21 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
         }
       }
       // ---- PV: OUT^T[d][q] += V^T x P -----------------------------------
-      // A-frags V^T[d = a*32 + col][key = st*16 + hi*8 + j] come from the
-      // NATURAL V image via ds_read_b64_tr_b16 (same additive-address
-      // derivation as flash_attn_bwd.hip DKV_TR_STEP, roles q->key)
-      const unsigned vtr_addr = vtr_base + (unsigned)(buf * KVB * D * 2);
-
-#define FWD_PV_STEP(a_, st_)                                               \
-      {                                                                    \
-        attn_u32x2 vl_, vh_;                                               \
-        asm volatile(                                                      \
-            "ds_read_b64_tr_b16 %0, %2 offset:%c3\n\t"                   \
-            "ds_read_b64_tr_b16 %1, %2 offset:%c4\n\t"                   \
-            "s_waitcnt lgkmcnt(0)"                                         \
-            : "=v"(vl_), "=v"(vh_)                                         \
-            : "v"(vtr_addr),                                               \
-              "i"((st_) * 16 * D * 2 + (a_) * 64),                         \
-              "i"((st_) * 16 * D * 2 + 4 * D * 2 + ((a_) ^ 1) * 64));      \
-        attn_u32x4 uv_ = {vl_.x, vl_.y, vh_.x, vh_.y};                     \
-        bf16x8 vf = __builtin_bit_cast(bf16x8, uv_);                       \
-        bf16x8 pf = *reinterpret_cast<const bf16x8*>(&pb[st_][0]);         \
-        oacc[a_] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(                \
-            vf, pf, oacc[a_], 0, 0, 0);                                    \
+#pragma unroll
+      for (int a = 0; a < NA; ++a) {
+#pragma unroll
+        for (int st = 0; st < 4; ++st) {
+          // A-frag: V^T[d = a*32 + col][key = st*16 + hi*8 + j]
+          const int d = a * 32 + col;
+          unsigned byte = (unsigned)d * (KVB * 2) + (st * 16 + hi * 8) * 2;
+          byte ^= (unsigned)(((d >> 3) & 7) << 4);
+          bf16x8 vf = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<const char*>(vbuf) + byte);
+          bf16x8 pf = *reinterpret_cast<const bf16x8*>(&pb[st][0]);
+          oacc[a] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pf, oacc[a],
+                                                            0, 0, 0);
+        }
       }
-#define FWD_PV_A(a_)                                                       \
-      FWD_PV_STEP(a_, 0);                                                  \
-      FWD_PV_STEP(a_, 1);                                                  \
-      FWD_PV_STEP(a_, 2);                                                  \
-      FWD_PV_STEP(a_, 3);
-      FWD_PV_A(0);
-      FWD_PV_A(1);
-      if constexpr (NA > 2) {
-        FWD_PV_A(2);
-        FWD_PV_A(3);
-      }
-#undef FWD_PV_A
-#undef FWD_PV_STEP
     }
 
     if (tile + 1 < ntiles) {
