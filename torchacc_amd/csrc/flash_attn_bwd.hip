@@ -200,6 +200,15 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
       }
       __syncthreads();
 
+      // per-wave skip: this wave's keys [key_b, key_b+31] have no valid q
+      // in the tile (causal diagonal / window-left bound) — staging and
+      // barriers are cooperative, so only the compute is guarded
+      bool wave_active = true;
+      if (CAUSAL && q0 + QT - 1 < key_b - shift) wave_active = false;
+      if (HAS_WINDOW && wl >= 0 && q0 > key_b + 31 - shift + wl)
+        wave_active = false;
+
+      if (wave_active) {
       // S = Q K^T and dP = dO V^T, both D[q=CROW][key=col]
       f32x16 s = f32x16(0.f);
       f32x16 dp = f32x16(0.f);
@@ -298,6 +307,7 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
         }
       }
 #undef DKV_TR_STEP
+      }  // wave_active
       __syncthreads();
     }
   }
