@@ -85,14 +85,11 @@ void fa_bwd_dkv_kernel_x(const short* __restrict__ dOut,
   // >=3 of d to spread write banks; reads (b128, 16-lane groups over
   // d = a*32+col) become conflict-free since (d&3, (d>>3)&3) is distinct
   // per lane
-  constexpr int TS = 32;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* q_lds = reinterpret_cast<short*>(smem);            // [QT][D] swz
   short* do_lds = q_lds + QT * D;                           // [QT][D] swz
-  short* qt_lds = do_lds + QT * D;                          // [D][TS]
-  short* dot_lds = qt_lds + D * TS;                         // [D][TS]
-  short* v_lds = dot_lds + D * TS;                          // [KVWG][D] swz
+  short* v_lds = do_lds + QT * D;                           // [KVWG][D] swz
   float* lse_lds = reinterpret_cast<float*>(v_lds + KVWG * D);
   float* del_lds = lse_lds + QT;
 
@@ -155,6 +152,13 @@ void fa_bwd_dkv_kernel_x(const short* __restrict__ dOut,
     __syncthreads();
   }
 
+  // per-lane constant address for the tr_b16 B-frag reads
+  const int m4 = (lane & 15) >> 2;
+  const int dl0 = ((lane >> 4) & 1) * 16 + (lane & 3) * 4;
+  const unsigned tr_addr = (unsigned)(size_t)q_lds +
+      (unsigned)((hi * 8 + m4) * (D * 2)) +
+      (((unsigned)(dl0 * 2)) ^ ((unsigned)(m4 << 4)));
+
   f32x16 dkacc[NA], dvacc[NA];
 #pragma unroll
   for (int a = 0; a < NA; ++a) {
@@ -196,15 +200,6 @@ void fa_bwd_dkv_kernel_x(const short* __restrict__ dOut,
               qv;
           *reinterpret_cast<s16x8*>(reinterpret_cast<char*>(do_lds) + byte) =
               dv8;
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            unsigned tb = (unsigned)(d0 + j) * (TS * 2) + row * 2;
-            tb ^= (unsigned)((((d0 + j) >> 3) & 3) << 4);
-            *reinterpret_cast<short*>(
-                reinterpret_cast<char*>(qt_lds) + tb) = qv[j];
-            *reinterpret_cast<short*>(
-                reinterpret_cast<char*>(dot_lds) + tb) = dv8[j];
-          }
         }
         for (int r = tid; r < QT; r += 256) {
           const int qrow = q0 + r;
@@ -267,7 +262,33 @@ void fa_bwd_dkv_kernel_x(const short* __restrict__ dOut,
           dp[r] = p * (dp[r] - del_q) * scale;
         }
       }
-      // A-frags over the q k-dim; step tp covers q rows 16tp..16tp+15
+      // B-frags via ds_read_b64_tr_b16 from the natural swizzled images
+      // (see flash_attn_bwd.hip DKV_TR_STEP for the address derivation)
+#define DKVX_TR_STEP(tp_, a_)                                              \
+      {                                                                    \
+        attn_u32x2 ql_, qh_, dl_, dh_;                                     \
+        asm volatile(                                                      \
+            "ds_read_b64_tr_b16 %0, %4 offset:%c5\n\t"                   \
+            "ds_read_b64_tr_b16 %1, %4 offset:%c6\n\t"                   \
+            "ds_read_b64_tr_b16 %2, %4 offset:%c7\n\t"                   \
+            "ds_read_b64_tr_b16 %3, %4 offset:%c8\n\t"                   \
+            "s_waitcnt lgkmcnt(0)"                                         \
+            : "=v"(ql_), "=v"(qh_), "=v"(dl_), "=v"(dh_)                   \
+            : "v"(tr_addr),                                                \
+              "i"((tp_) * 16 * D * 2 + (a_) * 64),                         \
+              "i"((tp_) * 16 * D * 2 + 4 * D * 2 + ((a_) ^ 1) * 64),       \
+              "i"(QT * D * 2 + (tp_) * 16 * D * 2 + (a_) * 64),            \
+              "i"(QT * D * 2 + (tp_) * 16 * D * 2 + 4 * D * 2 +            \
+                  ((a_) ^ 1) * 64));                                       \
+        attn_u32x4 uq_ = {ql_.x, ql_.y, qh_.x, qh_.y};                     \
+        attn_u32x4 ud_ = {dl_.x, dl_.y, dh_.x, dh_.y};                     \
+        bf16x8 qbf = __builtin_bit_cast(bf16x8, uq_);                      \
+        bf16x8 dob = __builtin_bit_cast(bf16x8, ud_);                      \
+        dvacc[a_] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(               \
+            pb, dob, dvacc[a_], 0, 0, 0);                                  \
+        dkacc[a_] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(               \
+            dsb, qbf, dkacc[a_], 0, 0, 0);                                 \
+      }
 #pragma unroll
       for (int tp = 0; tp < 2; ++tp) {
         unsigned pfr[4], dsfr[4];
@@ -275,25 +296,23 @@ void fa_bwd_dkv_kernel_x(const short* __restrict__ dOut,
         t12_pack_frag(dp, tp, dsfr);
         bf16x8 pb = *reinterpret_cast<const bf16x8*>(pfr);
         bf16x8 dsb = *reinterpret_cast<const bf16x8*>(dsfr);
-#pragma unroll
-        for (int a = 0; a < NA; ++a) {
-          // B-frags: X[q = 16tp + hi*8 + jj][d = a*32 + col] from the
-          // transposed tiles (contiguous b128, stride-40 rows)
-          const int d = a * 32 + col;
-          const int qoff = 16 * tp + hi * 8;
-          unsigned tb = (unsigned)d * (TS * 2) + qoff * 2;
-          tb ^= (unsigned)(((d >> 3) & 3) << 4);
-          bf16x8 dob = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<const char*>(dot_lds) + tb);
-          bf16x8 qbf = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<const char*>(qt_lds) + tb);
-          // dV += P^T dO ; dK += dS^T Q   (D[key=CROW][d=col])
-          dvacc[a] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              pb, dob, dvacc[a], 0, 0, 0);
-          dkacc[a] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              dsb, qbf, dkacc[a], 0, 0, 0);
+        if (tp == 0) {
+          DKVX_TR_STEP(0, 0);
+          DKVX_TR_STEP(0, 1);
+          if constexpr (NA > 2) {
+            DKVX_TR_STEP(0, 2);
+            DKVX_TR_STEP(0, 3);
+          }
+        } else {
+          DKVX_TR_STEP(1, 0);
+          DKVX_TR_STEP(1, 1);
+          if constexpr (NA > 2) {
+            DKVX_TR_STEP(1, 2);
+            DKVX_TR_STEP(1, 3);
+          }
         }
       }
+#undef DKVX_TR_STEP
       __syncthreads();
     }
   }
@@ -566,7 +585,7 @@ static void launch_fa_bwd_x(const torch::Tensor& dout, const torch::Tensor& q,
       alp, p_drop, rng_seed
 
   dim3 gkv((sk + 127) / 128, b * hk), bkv(256);
-  const int lds_kv = (2 * 32 * D + 2 * D * 32 + 128 * D) * 2 + 2 * 32 * 4;
+  const int lds_kv = (2 * 32 * D + 128 * D) * 2 + 2 * 32 * 4;
   dim3 gq((sq + 255) / 256, b * hq), bq(512);
   const int lds_q = 3 * 64 * D * 2;
 
