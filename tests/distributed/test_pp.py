@@ -202,3 +202,41 @@ def test_pp2_tied_embedding_matches_single_process():
         opt.zero_grad()
         base.append(float(loss))
     assert results[0] == pytest.approx(base, abs=1e-5)
+
+
+def _pp_infer_worker(rank, world, outdir):
+    import numpy as np
+    import os
+    import torchacc_amd as ta
+    from torchacc_amd.models import LlamaForCausalLM, llama_tiny
+    cfg = ta.Config()
+    cfg.dist.pp.size = world
+    cfg.dist.pp.num_micro_batches = 2
+    cfg.dist.pp.input_names = ["input_ids"]
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(llama_tiny())
+    model = ta.accelerate(model, config=cfg)
+    torch.manual_seed(9)
+    ids = torch.randint(0, 1024, (4, 16))
+    out = model(ids)  # inference schedule (no labels -> logits)
+    if rank == world - 1:
+        np.save(os.path.join(outdir, "pp_logits.npy"),
+                out.detach().float().numpy())
+
+
+def test_pp2_inference_matches_single_process(tmp_path):
+    """executor.forward (inference schedule): last stage's concatenated
+    logits must match the single-process model."""
+    run_multiprocess(_pp_infer_worker, world_size=2, args=(str(tmp_path),))
+    import numpy as np
+    got = np.load(tmp_path / "pp_logits.npy")
+
+    torch.manual_seed(0)
+    from torchacc_amd.models import LlamaForCausalLM, llama_tiny
+    model = LlamaForCausalLM(llama_tiny())
+    torch.manual_seed(9)
+    ids = torch.randint(0, 1024, (4, 16))
+    with torch.no_grad():
+        ref = model(ids).float().numpy()
+    assert got.shape == ref.shape
+    assert abs(got - ref).max() < 1e-4, abs(got - ref).max()
