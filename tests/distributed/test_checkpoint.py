@@ -110,3 +110,19 @@ def _optim_state_worker(rank, world, ckpt_dir):
 
 def test_full_optim_state_roundtrip():
     run_multiprocess(_optim_state_worker, world_size=2, args=("",))
+
+
+def test_ckpt_cli_entry(tmp_path):
+    """Console-script main() (consolidate_and_reshard_fsdp_ckpts) end to
+    end over a 2-rank checkpoint."""
+    ckpt_dir = str(tmp_path / "ckpt")
+    os.makedirs(ckpt_dir)
+    run_multiprocess(_save_worker, world_size=2, args=(ckpt_dir,))
+
+    from torchacc_amd.utils.consolidate_and_reshard_ckpts import main
+    out_dir = str(tmp_path / "cli_out")
+    main(["--ckpt_dir", ckpt_dir, "--ckpt_type", "model",
+          "--reshard_num", "2", "--output_dir", out_dir])
+    assert os.path.exists(os.path.join(out_dir, "consolidated_model.pth"))
+    assert len([f for f in os.listdir(out_dir)
+                if f.startswith("rank-") and f.endswith("-model.pth")]) == 2
