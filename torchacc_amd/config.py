@@ -8,9 +8,7 @@ Config :341) for a single eager PyTorch-ROCm backend: there is no 'lazy' vs
 kernels on the hot path.
 """
 from dataclasses import dataclass, field
-from typing import Any, Callable, Dict, List, Optional, Set, Union
-
-import torch
+from typing import Any, Callable, Dict, List, Optional, Set
 
 from .utils.logger import logger
 

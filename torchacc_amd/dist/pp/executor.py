@@ -11,7 +11,7 @@ intermediate stage's send list; during backward each stage accumulates the
 grad received from downstream into the grad it forwards upstream, plus any
 local autograd contribution if the value was also consumed here.
 """
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List
 
 import torch
 import torch.distributed as dist

@@ -12,7 +12,7 @@ activations downstream while receiving gradients from downstream — into one
 (the reference handled this with its even/odd op ordering,
 utils.py:403-408; on a chain the fused batch is the standard solution).
 """
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.distributed as dist

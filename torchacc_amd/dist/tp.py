@@ -15,8 +15,6 @@ column/row sharding with all-reduce on the TP group over xGMI
   compatibility (no-ops that explain the eager design).
 """
 import math
-from typing import Optional
-
 import torch
 import torch.distributed as dist
 import torch.nn as nn

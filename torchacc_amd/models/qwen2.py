@@ -14,7 +14,6 @@ from ..ops.cross_entropy import linear_cross_entropy
 from ..ops.flash_attn import flash_attn_xla
 from ..ops.rmsnorm import RMSNorm, fused_add_rms_norm
 from ..ops.rope import apply_rotary_pos_emb, build_rope_cache
-from ..ops.swiglu import swiglu
 from .llama import LlamaMLP
 
 

@@ -8,7 +8,6 @@ is a bug, not a degraded mode). CPU tensors always use the pure-torch
 reference math (used by the numerics tests as ground truth).
 """
 import importlib
-import os
 
 import torch
 

@@ -20,8 +20,6 @@ The reference's own ring/2D tests were skipped for correctness issues
 (test_context_parallel.py:104-109); this implementation is validated against
 single-device flash attention in tests/ops/test_context_parallel.py.
 """
-from typing import Optional
-
 import torch
 import torch.distributed as dist
 

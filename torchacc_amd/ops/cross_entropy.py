@@ -12,8 +12,6 @@ Replaces the reference's LigerCrossEntropyLoss / lce_forward integration
   GEMMs on hipBLASLt and the CE math on the CDNA4 kernel — on GPU the
   logits chunk stays bf16 end to end).
 """
-from typing import Optional
-
 import torch
 
 from ._backend import dispatch

@@ -15,13 +15,13 @@ flash_attn_varlen_position_ids_xla) on a single eager ROCm backend:
 - explicit cu_seqlens varlen (used by ring attention).
 
 The softmax log-sum-exp is returned in fp32 [b, h, s] exactly like FA2 so the
-context-parallel LSE merge math is unchanged. fp16/bf16 only. Dropout is
-accepted but only p=0.0 is supported by the CDNA4 kernels so far.
+context-parallel LSE merge math is unchanged. bf16 on the CDNA4 kernels
+(other dtypes/head dims run a composite fallback); alibi and dropout
+dispatch to the full-featured kernel family (csrc/flash_attn_extra_*.hip).
 """
 import math
 import threading
 from collections import deque
-from typing import Optional, Tuple
 
 import torch
 

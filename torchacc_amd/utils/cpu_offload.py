@@ -16,8 +16,7 @@ TransformerEngine-derived) natively for HIP streams:
 On CPU-only hosts the machinery degrades to a no-op (tensors stay put) so
 the same code paths are unit-testable without a GPU.
 """
-import functools
-from typing import Dict, List, Optional
+from typing import Dict, Optional
 
 import torch
 
