@@ -182,7 +182,6 @@ class PipeExecutor:
         self.micro_kwargs = split_microbatches(full_kwargs, self.num_micro)
         sched = create_scheduler("train", self.num_micro, self.stages,
                                  self.stage_id)
-        self._float_send_cache = {}
         self._exec_schedule_train(sched, output_fn)
         self._sync_tied_grads()
         return self._aggregate_total_loss()
