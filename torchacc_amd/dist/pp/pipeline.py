@@ -53,7 +53,16 @@ class PipelineParallel(ParallelModule):
 
         import torch.fx as fx
         if not isinstance(model, fx.GraphModule):
-            gm = trace(model, pp_cfg.input_names)
+            try:
+                gm = trace(model, pp_cfg.input_names)
+            except Exception as e:
+                raise RuntimeError(
+                    "pipeline parallelism requires an fx-traceable model; "
+                    f"tracing {type(model).__name__} failed ({e}). "
+                    "transformers >= 5 removed its HF fx tracer, so HF "
+                    "models cannot be pipeline-split — use the native "
+                    "model families (torchacc_amd.models) for PP, or "
+                    "FSDP/DP/CP which need no tracing") from e
         else:
             gm = model
         sp = _preprocess_split_points(model, pp_cfg.split_points)

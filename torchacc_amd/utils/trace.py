@@ -78,9 +78,15 @@ def trace(model: torch.nn.Module,
     except ImportError:
         is_hf = False
     if is_hf:
-        from transformers.utils.fx import HFTracer, symbolic_trace as hf_st
-        gm = hf_st(model, input_names=input_names)
-        return gm
+        # transformers < 5 shipped an HF-aware tracer; transformers >= 5
+        # removed transformers.utils.fx, so fall through to the generic
+        # tracer (works for models whose forward branches are static given
+        # the concrete defaults)
+        try:
+            from transformers.utils.fx import symbolic_trace as hf_st
+            return hf_st(model, input_names=input_names)
+        except ImportError:
+            pass
     concrete = _concrete_args_from_signature(model, input_names)
     tracer = _Tracer()
     graph = tracer.trace(model, concrete_args=concrete or None)
