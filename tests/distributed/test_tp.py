@@ -89,3 +89,43 @@ def test_column_row_pair():
         rank, ok_fwd, ok_bwd = q.get()
         assert ok_fwd, f"rank {rank} forward mismatch"
         assert ok_bwd, f"rank {rank} backward mismatch"
+
+
+def _tp_fsdp_worker(rank, world, q):
+    """tp2 x fsdp2 on 4 ranks: TP shards heads/MLP inside each FSDP
+    replica group; losses must agree across all ranks and decrease."""
+    import torchacc_amd as ta
+    from torchacc_amd.models import LlamaForCausalLM, llama_tiny
+    cfg = ta.Config()
+    cfg.dist.tp.size = 2
+    cfg.dist.fsdp.size = 2
+    cfg.dist.fsdp.wrap_layer_cls = {"LlamaDecoderLayer"}
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(llama_tiny())
+    model = ta.accelerate(model, config=cfg)
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    torch.manual_seed(42)  # same data everywhere
+    data = [torch.randint(0, 1024, (2, 32)) for _ in range(4)]
+    losses = []
+    for i in range(4):
+        ids = data[i]
+        loss = model(ids, labels=ids)
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        losses.append(float(loss))
+    q.put((rank, losses))
+
+
+def test_tp2_fsdp2_compose():
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    run_multiprocess(_tp_fsdp_worker, world_size=4, args=(q,))
+    results = {}
+    for _ in range(4):
+        r, losses = q.get()
+        results[r] = losses
+    for r in range(1, 4):
+        assert results[r] == pytest.approx(results[0], abs=1e-4), r
+    assert results[0][-1] < results[0][0]
