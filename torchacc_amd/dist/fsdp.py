@@ -307,6 +307,12 @@ class FullyShardedDataParallel(ParallelModule):
         self._order_final = False
         self._backward_pending = 0
         self._sync_scheduled = False
+        if wrap_cls and len(self.units) == (1 if self.root_unit else 0):
+            logger.warning(
+                "FSDP: wrap_layer_cls %s matched no modules — the whole "
+                "model is ONE flat-param unit (no per-layer gather overlap)."
+                " Check the class names against model.named_modules()",
+                sorted(wrap_cls))
         if self.mesh.global_rank == 0:
             tot = sum(u.total_numel for u in self.units)
             logger.info("FSDP: %d units, %.1fM params, shard dtype %s, ws=%d",
