@@ -154,3 +154,42 @@ def test_no_sync_grad_accumulation():
     for _ in range(2):
         rank, ok = q.get()
         assert ok, f"rank {rank} no_sync accumulation mismatch"
+
+
+def _clip_worker(rank, world, q):
+    import torchacc_amd as ta
+    cfg = ta.Config()
+    cfg.dist.fsdp.size = world
+    cfg.dist.fsdp.wrap_layer_cls = {"LlamaDecoderLayer"}
+    model = _make_model()
+    wrapped = ta.accelerate(model, config=cfg)
+    torch.manual_seed(42)
+    ids = torch.randint(0, 1024, (2, 32))
+    wrapped(ids, labels=ids).backward()
+    total = wrapped.clip_grad_norm_(0.1)
+    # after clipping, recompute the global shard norm: must equal max_norm
+    fsdp = wrapped.fsdp_wrapper
+    import torch.distributed as dist
+    local = sum(u.shard.grad.float().pow(2).sum() for u in fsdp.units
+                if u.shard.grad is not None)
+    dist.all_reduce(local)
+    q.put((rank, float(total), float(local.sqrt())))
+
+
+def test_clip_grad_norm_matches_single_process():
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    run_multiprocess(_clip_worker, world_size=2, args=(q,))
+    results = [q.get() for _ in range(2)]
+    # pre-clip norm identical across ranks and equals the single-process one
+    assert results[0][1] == pytest.approx(results[1][1], rel=1e-6)
+    model = _make_model()
+    torch.manual_seed(42)
+    ids = torch.randint(0, 1024, (2, 32))
+    model(ids, labels=ids).backward()
+    ref = torch.nn.utils.clip_grad_norm_(model.parameters(), 0.1)
+    assert results[0][1] == pytest.approx(float(ref), rel=1e-3)
+    # post-clip global norm == max_norm
+    for _, _, post in results:
+        assert post == pytest.approx(0.1, rel=1e-3)
