@@ -62,6 +62,16 @@ def llama_3_8b(**kw) -> "LlamaConfig":
     return LlamaConfig(**base)
 
 
+def llama_3_70b(**kw) -> "LlamaConfig":
+    """Llama-3 70B: GQA (8 kv heads), 128k vocab, rope theta 500k."""
+    base = dict(
+        vocab_size=128256, hidden_size=8192, intermediate_size=28672,
+        num_hidden_layers=80, num_attention_heads=64, num_key_value_heads=8,
+        max_position_embeddings=8192, rope_theta=500000.0)
+    base.update(kw)
+    return LlamaConfig(**base)
+
+
 def llama_tiny(**kw) -> "LlamaConfig":
     """4-layer toy config (driver config #1: tiny DP=1 CPU plumbing)."""
     base = dict(
