@@ -1,7 +1,7 @@
-"""Standalone PP smoke: 2-stage pipeline with gradient checkpointing on the
-stage modules (reference tests/standalone/pipeline.py).
+"""Standalone PP smoke: 4-stage pipeline with gradient checkpointing on
+the stage layers (reference tests/standalone/pipeline.py ran 4 stages).
 
-torchrun --nproc-per-node 2 --master-addr 127.0.0.1 \
+torchrun --nproc-per-node 4 --master-addr 127.0.0.1 \
     tests/standalone/pipeline.py
 """
 import os
@@ -22,6 +22,8 @@ def main():
     cfg.dist.pp.size = world
     cfg.dist.pp.num_micro_batches = 4
     cfg.dist.pp.input_names = ["input_ids", "labels"]
+    cfg.memory.gc = True
+    cfg.memory.gc_cls = {"LlamaDecoderLayer"}
     torch.manual_seed(0)
     model = LlamaForCausalLM(llama_tiny())
     model = ta.accelerate(model, config=cfg)
