@@ -172,3 +172,26 @@ def test_llama_gqa_forward_backward():
     loss = model(ids, labels=ids)
     loss.backward()
     assert torch.isfinite(loss)
+
+
+def test_patch_qwen_model_rebinds_remote_flash_attn():
+    """patch_qwen_model points a remote-code module's flash-attn symbols at
+    this framework's ops (reference regex-rewrote source; we rebind)."""
+    import sys
+    import types
+    import torch
+    from torchacc_amd import patch_qwen_model
+    from torchacc_amd.ops.flash_attn import flash_attn_varlen_func
+
+    fake = types.ModuleType("remote_qwen_modeling")
+    fake.flash_attn_unpadded_func = lambda *a, **k: None
+    sys.modules["remote_qwen_modeling"] = fake
+    try:
+        class FakeQwen(torch.nn.Module):
+            pass
+        FakeQwen.__module__ = "remote_qwen_modeling"
+        m = FakeQwen()
+        assert patch_qwen_model(m) is True
+        assert fake.flash_attn_unpadded_func is flash_attn_varlen_func
+    finally:
+        del sys.modules["remote_qwen_modeling"]
