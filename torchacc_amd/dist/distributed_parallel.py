@@ -73,6 +73,15 @@ class DistributedParallel(ParallelModule):
                                               max_grad_norm)
 
     # FSDP optim-state passthroughs (reference distributed_parallel.py:63-111)
+    def no_sync(self):
+        """Gradient-accumulation context: skip grad reduction inside; the
+        first synchronized backward after the context reduces the
+        accumulated total (FSDP engine; no-op without FSDP)."""
+        if self.fsdp_wrapper is not None:
+            return self.fsdp_wrapper.no_sync()
+        import contextlib
+        return contextlib.nullcontext()
+
     def sharded_state_dict(self):
         assert self.fsdp_wrapper is not None
         return self.fsdp_wrapper.sharded_state_dict()
