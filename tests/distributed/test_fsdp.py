@@ -193,3 +193,31 @@ def test_clip_grad_norm_matches_single_process():
     # post-clip global norm == max_norm
     for _, _, post in results:
         assert post == pytest.approx(0.1, rel=1e-3)
+
+
+def _sync_states_worker(rank, world, q):
+    import torchacc_amd as ta
+    cfg = ta.Config()
+    cfg.dist.fsdp.size = world
+    cfg.dist.fsdp.wrap_layer_cls = {"LlamaDecoderLayer"}
+    cfg.dist.fsdp.sync_module_states = True
+    torch.manual_seed(rank * 97 + 1)  # DIFFERENT init per rank
+    model = _make_model(seed=rank * 97 + 1)
+    wrapped = ta.accelerate(model, config=cfg)
+    full = wrapped.full_state_dict()
+    sig = float(sum(v.float().sum() for v in full.values()))
+    q.put((rank, sig))
+
+
+def test_sync_module_states_broadcasts_rank0():
+    """sync_module_states=True must make all ranks' full state identical
+    (broadcast from the group source) despite different local inits."""
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    run_multiprocess(_sync_states_worker, world_size=2, args=(q,))
+    sigs = {}
+    for _ in range(2):
+        r, s = q.get()
+        sigs[r] = s
+    assert sigs[0] == pytest.approx(sigs[1], rel=1e-6), sigs
