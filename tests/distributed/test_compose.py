@@ -88,3 +88,57 @@ def test_gc_cnt_limits_wrapping():
     loss = model(ids, labels=ids)
     loss.backward()
     assert torch.isfinite(loss)
+
+
+def _hybrid_worker(rank, world, q):
+    """fsdp2 x dp2 hybrid shard: shards inside each DP replica, gradient
+    all-reduce across replicas."""
+    import torchacc_amd as ta
+    from torchacc_amd.models import LlamaForCausalLM, llama_tiny
+    cfg = ta.Config()
+    cfg.dist.fsdp.size = 2
+    cfg.dist.dp.size = 2
+    cfg.dist.fsdp.wrap_layer_cls = {"LlamaDecoderLayer"}
+    cfg.dist.topology = ["dp", "fsdp", "pp", "tp"]
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(llama_tiny())
+    model = ta.accelerate(model, config=cfg)
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    torch.manual_seed(42)  # same data everywhere -> parity with 1-proc
+    data = [torch.randint(0, 1024, (2, 32)) for _ in range(3)]
+    losses = []
+    for ids in data:
+        loss = model(ids, labels=ids)
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        losses.append(float(loss))
+    q.put((rank, losses))
+
+
+def test_fsdp2_dp2_hybrid():
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    run_multiprocess(_hybrid_worker, world_size=4, args=(q,))
+    results = {}
+    for _ in range(4):
+        r, losses = q.get()
+        results[r] = losses
+    for r in range(1, 4):
+        assert results[r] == pytest.approx(results[0], abs=1e-5), r
+
+    from torchacc_amd.models import LlamaForCausalLM, llama_tiny
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(llama_tiny())
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    torch.manual_seed(42)
+    data = [torch.randint(0, 1024, (2, 32)) for _ in range(3)]
+    base = []
+    for ids in data:
+        loss = model(ids, labels=ids)
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        base.append(float(loss))
+    assert results[0] == pytest.approx(base, abs=5e-3)
