@@ -195,3 +195,17 @@ def test_patch_qwen_model_rebinds_remote_flash_attn():
         assert fake.flash_attn_unpadded_func is flash_attn_varlen_func
     finally:
         del sys.modules["remote_qwen_modeling"]
+
+
+def test_bucketing_pad_value_dict():
+    """pad_value_dict controls the fill per key (labels -100 so CE ignores
+    the padding)."""
+    from torchacc_amd.async_loader import AsyncLoader
+    data = [{"input_ids": torch.ones(2, 100, dtype=torch.long),
+             "labels": torch.ones(2, 100, dtype=torch.long)}]
+    loader = AsyncLoader(data, "cpu", buckets=[128],
+                         pad_value_dict={"labels": -100})
+    batch = next(iter(loader))
+    assert batch["input_ids"].shape[-1] == 128
+    assert (batch["input_ids"][:, 100:] == 0).all()
+    assert (batch["labels"][:, 100:] == -100).all()
