@@ -1,0 +1,138 @@
+"""Round-2 correctness fixes (advisor findings + syncfree AMP contract).
+
+- gradient-checkpoint wrappers must not leak '_checkpoint_wrapped_module.'
+  into state dicts or FSDP shard metadata (gc-on and gc-off checkpoints are
+  interchangeable, matching the reference checkpoint format);
+- micro-batch splitting requires divisibility (uneven chunks would be
+  over-weighted by the executor's 1/num scaling);
+- the HF MLP kernel patch honors config.hidden_act;
+- ops.AdamW implements torch's _step_supports_amp_scaling contract
+  (device-side grad_scale/found_inf; reference syncfree semantics,
+  utils/patch.py:55-57).
+"""
+import pytest
+import torch
+
+from tests.utils.distributed import run_multiprocess
+
+
+def test_checkpoint_wrapper_state_dict_prefix():
+    from torchacc_amd.utils.checkpoint import gradient_checkpoint
+
+    torch.manual_seed(0)
+
+    def make():
+        return torch.nn.Sequential(
+            torch.nn.Linear(8, 8), torch.nn.Linear(8, 8))
+
+    plain = make()
+    wrapped = make()
+    wrapped.load_state_dict(plain.state_dict())
+    gradient_checkpoint(wrapped, gc_cls={"Linear"})
+    sd = wrapped.state_dict()
+    assert set(sd.keys()) == set(plain.state_dict().keys())
+    for k in sd:
+        assert "_checkpoint_wrapped_module" not in k
+    # a clean (gc-off) state dict loads into the gc-wrapped model
+    wrapped.load_state_dict(plain.state_dict())
+    # and the gc-wrapped state dict loads into a clean model
+    plain.load_state_dict(sd)
+
+
+def _gc_shard_names_worker(rank, world):
+    import torchacc_amd as ta
+    from torchacc_amd.models import LlamaForCausalLM, llama_tiny
+    cfg = ta.Config()
+    cfg.dist.fsdp.size = world
+    cfg.dist.fsdp.wrap_layer_cls = {"LlamaDecoderLayer"}
+    cfg.memory.gc = True
+    cfg.memory.gc_cls = {"LlamaDecoderLayer"}
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(llama_tiny())
+    clean_keys = {k for k in model.state_dict().keys()
+                  if not k.startswith("rope_")}
+    wrapped = ta.accelerate(model, config=cfg)
+    meta = wrapped.fsdp_wrapper.shard_metadata()
+    names = {p["name"] for u in meta["units"] for p in u["params"]}
+    for n in names:
+        assert "_checkpoint_wrapped_module" not in n, n
+    assert names == clean_keys, names ^ clean_keys
+    full = wrapped.full_state_dict()
+    for k in full:
+        assert "_checkpoint_wrapped_module" not in k, k
+
+
+def test_fsdp_gc_shard_metadata_names_clean():
+    run_multiprocess(_gc_shard_names_worker, world_size=2)
+
+
+def test_microbatch_divisibility_required():
+    from torchacc_amd.dist.pp.microbatch import split_microbatches
+    ok = split_microbatches({"x": torch.zeros(8, 3)}, 4)
+    assert len(ok) == 4 and all(m["x"].shape[0] == 2 for m in ok)
+    with pytest.raises(AssertionError):
+        split_microbatches({"x": torch.zeros(6, 3)}, 4)
+
+
+def test_hf_mlp_patch_honors_hidden_act():
+    transformers = pytest.importorskip("transformers")
+    from transformers.models.llama.configuration_llama import LlamaConfig
+    from transformers.models.llama.modeling_llama import LlamaMLP
+
+    from torchacc_amd.utils.patch import apply_fused_kernel_patches
+
+    apply_fused_kernel_patches()
+    torch.manual_seed(0)
+    x = torch.randn(2, 4, 32)
+
+    gelu_cfg = LlamaConfig(hidden_size=32, intermediate_size=64,
+                           hidden_act="gelu")
+    mlp = LlamaMLP(gelu_cfg)
+    got = mlp(x)
+    want = mlp.down_proj(
+        torch.nn.functional.gelu(mlp.gate_proj(x)) * mlp.up_proj(x))
+    assert torch.allclose(got, want, atol=1e-6), \
+        "gelu MLP must not be silently replaced by silu"
+
+    silu_cfg = LlamaConfig(hidden_size=32, intermediate_size=64,
+                           hidden_act="silu")
+    mlp2 = LlamaMLP(silu_cfg)
+    got2 = mlp2(x)
+    want2 = mlp2.down_proj(
+        torch.nn.functional.silu(mlp2.gate_proj(x)) * mlp2.up_proj(x))
+    assert torch.allclose(got2, want2, atol=1e-5)
+
+
+def test_adamw_step_supports_amp_scaling():
+    from torchacc_amd.ops.adamw import AdamW
+
+    assert AdamW._step_supports_amp_scaling is True
+
+    torch.manual_seed(0)
+    p_ref = torch.nn.Parameter(torch.randn(16))
+    p_scaled = torch.nn.Parameter(p_ref.detach().clone())
+    g = torch.randn(16)
+
+    opt_ref = AdamW([p_ref], lr=1e-2)
+    p_ref.grad = g.clone()
+    opt_ref.step()
+
+    # grads still carry the loss scale; the GradScaler contract hands the
+    # optimizer grad_scale/found_inf attributes and the step unscales itself
+    scale = 1024.0
+    opt = AdamW([p_scaled], lr=1e-2)
+    p_scaled.grad = g * scale
+    opt.grad_scale = torch.tensor(scale)
+    opt.found_inf = torch.zeros(())
+    opt.step()
+    assert torch.allclose(p_scaled.detach(), p_ref.detach(), atol=1e-6)
+
+
+def test_adamw_skips_on_found_inf():
+    from torchacc_amd.ops.adamw import AdamW
+    p = torch.nn.Parameter(torch.randn(8))
+    before = p.detach().clone()
+    opt = AdamW([p], lr=1.0)
+    p.grad = torch.randn(8)
+    opt.step(found_inf=torch.ones(()))
+    assert torch.equal(p.detach(), before)

@@ -23,10 +23,15 @@ class GradScaler(torch.amp.GradScaler):
     def _unscale_grads_(self, optimizer, inv_scale, found_inf,
                         allow_fp16=False):
         # pure-fp16 training keeps fp32 masters inside the fused AdamW, so
-        # unscaling fp16 grads directly is safe (torch's default forbids it
-        # because its optimizers would update fp16 params in place)
+        # unscaling fp16 grads directly is safe FOR THAT OPTIMIZER only;
+        # torch's default forbids it because plain optimizers would update
+        # fp16 params in place from fp16 grads
+        from .ops.adamw import AdamW as FusedAdamW
+        allow = allow_fp16 or (
+            isinstance(optimizer, FusedAdamW) and
+            getattr(optimizer, "use_master_weights", False))
         out = super()._unscale_grads_(optimizer, inv_scale, found_inf,
-                                      allow_fp16=True)
+                                      allow_fp16=allow)
         if self._pp_group is not None and dist.is_initialized() and \
                 dist.get_world_size(self._pp_group) > 1:
             for v in out.values():

@@ -116,7 +116,19 @@ class AsyncLoader:
                 raise item
             batch, ev = item
             if ev is not None:
-                torch.cuda.current_stream().wait_event(ev)
+                cur = torch.cuda.current_stream()
+                cur.wait_event(ev)
+                # the device tensors were allocated on the side h2d stream;
+                # mark them in-use by the consumer stream so the caching
+                # allocator cannot recycle their blocks for the next batch's
+                # copy while compute kernels still read them
+
+                def _claim(t: torch.Tensor):
+                    if t.is_cuda:
+                        t.record_stream(cur)
+                    return t
+
+                recursively_apply(_claim, batch)
             yield batch
 
 

@@ -34,6 +34,7 @@ from typing import Dict, List, Optional, Tuple
 import torch
 import torch.distributed as dist
 
+from ..utils.checkpoint import strip_checkpoint_prefix as _clean
 from ..utils.logger import logger
 from .backend import get_comm_stream
 from .parallel_module import ParallelModule
@@ -550,13 +551,16 @@ class FullyShardedDataParallel(ParallelModule):
             "pad_multiple": PAD_MULTIPLE,
             "flat_dtype": str(self.flat_dtype),
             "units": units_meta,
-            "buffers": [k for k, _ in self.model.named_buffers()],
+            "buffers": [_clean(k) for k, _ in self.model.named_buffers()],
         }
 
     def _param_full_name(self, owner: torch.nn.Module, attr: str) -> str:
         for name, mod in self.model.named_modules():
             if mod is owner:
-                return f"{name}.{attr}" if name else attr
+                full = f"{name}.{attr}" if name else attr
+                # accelerate() applies gradient checkpointing after FSDP;
+                # keep ckpt layouts identical with and without memory.gc
+                return _clean(full)
         return attr
 
     def sharded_state_dict(self) -> dict:
@@ -564,7 +568,7 @@ class FullyShardedDataParallel(ParallelModule):
         the reference's rank-*-of-*-model.pth payload shape."""
         model_sd = {u.name: u.shard.detach().cpu() for u in self.units}
         for k, b in self.model.named_buffers():
-            model_sd[f"__buffer__.{k}"] = b.detach().cpu()
+            model_sd[f"__buffer__.{_clean(k)}"] = b.detach().cpu()
         return {"model": model_sd, "shard_metadata": self.shard_metadata()}
 
     def load_sharded_state_dict(self, payload: dict):
@@ -573,7 +577,7 @@ class FullyShardedDataParallel(ParallelModule):
             for u in self.units:
                 u.shard.copy_(sd[u.name].to(u.shard.device, u.shard.dtype))
             for k, b in self.model.named_buffers():
-                key = f"__buffer__.{k}"
+                key = f"__buffer__.{_clean(k)}"
                 if key in sd:
                     b.copy_(sd[key].to(b.device, b.dtype))
 
@@ -595,5 +599,5 @@ class FullyShardedDataParallel(ParallelModule):
                 if not was:
                     u.reshard()
             for k, b in self.model.named_buffers():
-                out[k] = b.detach().cpu()
+                out[_clean(k)] = b.detach().cpu()
         return out

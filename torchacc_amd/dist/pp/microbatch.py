@@ -18,9 +18,10 @@ def split_microbatches(kwargs: Dict, num_micro_batches: int) -> List[Dict]:
     chunked = {}
     for k, v in kwargs.items():
         if isinstance(v, torch.Tensor):
-            assert v.shape[0] % num_micro_batches == 0 or \
-                v.shape[0] >= num_micro_batches, \
-                (f"batch dim {v.shape[0]} of '{k}' cannot be split into "
+            # uneven chunks would be over-weighted: the executor scales each
+            # micro-batch loss by 1/num_micro_batches
+            assert v.shape[0] % num_micro_batches == 0, \
+                (f"batch dim {v.shape[0]} of '{k}' must be divisible by "
                  f"{num_micro_batches} micro-batches")
             chunked[k] = list(v.chunk(num_micro_batches, dim=0))
         else:

@@ -18,6 +18,12 @@ from ._backend import dispatch
 
 class AdamW(torch.optim.Optimizer):
 
+    # torch.amp.GradScaler.step() sees this and hands us device-side
+    # ``grad_scale``/``found_inf`` tensors instead of host-syncing on
+    # found_inf.item() — the fully syncfree fp16 path (reference
+    # torch_xla.amp.syncfree semantics, utils/patch.py:55-57)
+    _step_supports_amp_scaling = True
+
     def __init__(self, params, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
                  weight_decay=1e-2, use_master_weights: bool = True):
         defaults = dict(lr=lr, betas=betas, eps=eps,
@@ -31,6 +37,16 @@ class AdamW(torch.optim.Optimizer):
         if closure is not None:
             with torch.enable_grad():
                 loss = closure()
+        # attributes installed by torch.amp.GradScaler.step (see
+        # _step_supports_amp_scaling): grad_scale is the CURRENT scale when
+        # grads are still scaled (None once unscale_ ran); found_inf is the
+        # device-side overflow flag
+        if found_inf is None:
+            found_inf = getattr(self, "found_inf", None)
+        grad_scale = getattr(self, "grad_scale", None)
+        inv_scale = None
+        if grad_scale is not None:
+            inv_scale = grad_scale.double().reciprocal().float()
         for group in self.param_groups:
             beta1, beta2 = group["betas"]
             lr = group["lr"]
@@ -73,6 +89,9 @@ class AdamW(torch.optim.Optimizer):
                     g if g.dtype == p.dtype else g.to(p.dtype)
                     for g, p in zip(grads, params)
                 ]
+                if inv_scale is not None:
+                    # grads still carry the loss scale: unscale device-side
+                    grads = torch._foreach_mul(grads, inv_scale)
                 ext.fused_adamw(params, grads, exp_avgs, exp_avg_sqs,
                                 master_list, steps, fi, lr, beta1, beta2,
                                 eps, wd)
@@ -82,6 +101,8 @@ class AdamW(torch.optim.Optimizer):
                 for p, g, m, v, mw, t in zip(params, grads, exp_avgs,
                                              exp_avg_sqs, masters, steps):
                     gf = g.float()
+                    if inv_scale is not None:
+                        gf = gf * inv_scale
                     m.mul_(beta1).add_(gf, alpha=1 - beta1)
                     v.mul_(beta2).addcmul_(gf, gf, value=1 - beta2)
                     bc1 = 1 - beta1 ** t

@@ -281,6 +281,26 @@ def _ref_fa_backward(dout, q, k, v, out, lse, softmax_scale, causal, window,
 # user wrappers (reference-compatible names)
 # ---------------------------------------------------------------------------
 
+_warned_fp16 = [False]
+
+
+def _cast_fp16(q, k, v):
+    """fp16 inputs on GPU run the bf16 MFMA kernels through an explicit
+    (warned, autograd-tracked) cast — never the fp32 composite, which would
+    materialize [b,h,s,s] scores (reference accepts fp16 natively,
+    ops/flash_attn.py:324-325). Returns (q, k, v, needs_cast_back)."""
+    if q.is_cuda and q.dtype == torch.float16:
+        if not _warned_fp16[0]:
+            _warned_fp16[0] = True
+            from ..utils.logger import logger
+            logger.warning(
+                "flash-attention fp16 inputs: computing in bf16 on the "
+                "CDNA4 MFMA kernels (output cast back to fp16)")
+        return q.to(torch.bfloat16), k.to(torch.bfloat16), \
+            v.to(torch.bfloat16), True
+    return q, k, v, False
+
+
 def flash_attn_xla(q, k, v, dropout_p=0.0, softmax_scale=None, causal=False,
                    window_size=(-1, -1), alibi_slopes=None,
                    deterministic=False, return_attn_probs=False):
@@ -289,9 +309,12 @@ def flash_attn_xla(q, k, v, dropout_p=0.0, softmax_scale=None, causal=False,
     Name kept from the reference for drop-in compatibility; runs on the
     CDNA4 HIP kernel (there is no XLA here).
     """
+    q, k, v, back = _cast_fp16(q, k, v)
     out, lse = FlashAttnFunc.apply(q, k, v, dropout_p, softmax_scale, causal,
                                    window_size, alibi_slopes, deterministic,
                                    None, None)
+    if back:
+        out = out.to(torch.float16)
     if return_attn_probs:
         return out, lse, None
     return out
@@ -314,9 +337,12 @@ def flash_attn_varlen_xla(q, k, v, attention_mask=None, dropout_p=0.0,
         k_lens = attention_mask.to(torch.int32).sum(-1).to(torch.int32)
         if q.shape[1] == k.shape[1]:
             q_lens = k_lens
+    q, k, v, back = _cast_fp16(q, k, v)
     out, lse = FlashAttnFunc.apply(q, k, v, dropout_p, softmax_scale, causal,
                                    window_size, alibi_slopes, deterministic,
                                    q_lens, k_lens)
+    if back:
+        out = out.to(torch.float16)
     if return_attn_probs:
         return out, lse, None
     return out
@@ -345,10 +371,13 @@ def flash_attn_varlen_position_ids_xla(q, k, v, position_ids, dropout_p=0.0,
     assert q.shape[0] == 1, "position-ids varlen requires batch size 1"
     cu = position_ids_to_cu_seqlens(position_ids)
     max_len = int((cu[1:] - cu[:-1]).max())
+    q, k, v, back = _cast_fp16(q, k, v)
     out, lse = FlashAttnVarlenFunc.apply(
         q.squeeze(0), k.squeeze(0), v.squeeze(0), cu, cu, max_len, max_len,
         dropout_p, softmax_scale, causal, window_size, deterministic)
     out = out.unsqueeze(0)
+    if back:
+        out = out.to(torch.float16)
     if return_attn_probs:
         return out, lse, None
     return out
@@ -524,10 +553,13 @@ def flash_attn_varlen_func(q, k, v, cu_seqlens_q, cu_seqlens_k, max_seqlen_q,
                            causal=False, window_size=(-1, -1),
                            alibi_slopes=None, deterministic=False,
                            return_attn_probs=False):
+    q, k, v, back = _cast_fp16(q, k, v)
     out, lse = FlashAttnVarlenFunc.apply(q, k, v, cu_seqlens_q, cu_seqlens_k,
                                          max_seqlen_q, max_seqlen_k,
                                          dropout_p, softmax_scale, causal,
                                          window_size, deterministic)
+    if back:
+        out = out.to(torch.float16)
     if return_attn_probs:
         return out, lse, None
     return out

@@ -13,12 +13,40 @@ from torch.utils.checkpoint import checkpoint
 from .logger import logger
 
 
+_CKPT_PREFIX = "_checkpoint_wrapped_module."
+
+
+def strip_checkpoint_prefix(name: str) -> str:
+    """Remove every '_checkpoint_wrapped_module.' segment from a dotted
+    parameter/buffer name so state dicts are identical with and without
+    gradient checkpointing (and match the reference checkpoint format)."""
+    return name.replace(_CKPT_PREFIX, "")
+
+
 class CheckpointWrapper(torch.nn.Module):
 
     def __init__(self, mod: torch.nn.Module, selective_attn: bool = False):
         super().__init__()
         self._checkpoint_wrapped_module = mod
         self._selective_attn = selective_attn
+        self._register_state_dict_hook(self._strip_prefix_hook)
+        self._register_load_state_dict_pre_hook(
+            self._add_prefix_hook, with_module=True)
+
+    @staticmethod
+    def _strip_prefix_hook(module, state_dict, prefix, local_metadata):
+        wrapped = prefix + _CKPT_PREFIX
+        for key in list(state_dict.keys()):
+            if key.startswith(wrapped):
+                state_dict[prefix + key[len(wrapped):]] = state_dict.pop(key)
+        return state_dict
+
+    @staticmethod
+    def _add_prefix_hook(module, state_dict, prefix, *args):
+        wrapped = prefix + _CKPT_PREFIX
+        for key in list(state_dict.keys()):
+            if key.startswith(prefix) and not key.startswith(wrapped):
+                state_dict[wrapped + key[len(prefix):]] = state_dict.pop(key)
 
     def forward(self, *args, **kwargs):
         if torch.is_grad_enabled():

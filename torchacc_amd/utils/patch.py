@@ -140,7 +140,18 @@ def _patched_rmsnorm_forward(self, hidden_states):
     return rms_norm(hidden_states, self.weight, self.variance_epsilon)
 
 
+def _is_silu(act) -> bool:
+    if isinstance(act, torch.nn.SiLU):
+        return True
+    name = getattr(act, "__name__", type(act).__name__).lower()
+    return name in ("silu", "swish", "silu_", "silukernel")
+
+
 def _patched_mlp_forward(self, x):
+    act = getattr(self, "act_fn", None)
+    if act is not None and not _is_silu(act):
+        # honor config.hidden_act: only the SiLU gate is fused
+        return self.down_proj(act(self.gate_proj(x)) * self.up_proj(x))
     from ..ops.swiglu import swiglu
     return self.down_proj(swiglu(self.gate_proj(x), self.up_proj(x)))
 
