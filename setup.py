@@ -24,6 +24,7 @@ sources = [
         "flash_attn_varlen.hip",
         "flash_attn_extra_fwd.hip",
         "flash_attn_extra_bwd.hip",
+        "gemm.hip",
     )
 ]
 
@@ -35,6 +36,7 @@ setup(
         CUDAExtension(
             name="torchacc_amd._C",
             sources=sources,
+            libraries=["hipblaslt"],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17"],

@@ -66,6 +66,15 @@ std::vector<torch::Tensor> fa_varlen_backward(
     torch::Tensor out, torch::Tensor lse, torch::Tensor bounds,
     double softmax_scale, bool causal);
 
+// gemm.hip (hipBLASLt tuned GEMM)
+std::vector<int64_t> lt_gemm_candidates(int64_t m, int64_t n, int64_t k,
+                                        bool ta, bool tb,
+                                        int64_t max_workspace_mb);
+torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, bool ta, bool tb,
+                      int64_t algo_index,
+                      c10::optional<torch::Tensor> out_opt);
+std::string lt_gemm_algo_name(int64_t algo_index);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "torchacc_amd CDNA4 (gfx950) kernels";
   m.def("rmsnorm_forward", &rmsnorm_forward);
@@ -82,4 +91,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fa_backward", &fa_backward);
   m.def("fa_varlen_forward", &fa_varlen_forward);
   m.def("fa_varlen_backward", &fa_varlen_backward);
+  m.def("lt_gemm_candidates", &lt_gemm_candidates);
+  m.def("lt_gemm", &lt_gemm, pybind11::arg("a"), pybind11::arg("b"),
+        pybind11::arg("ta"), pybind11::arg("tb"),
+        pybind11::arg("algo_index") = -1,
+        pybind11::arg("out") = pybind11::none());
+  m.def("lt_gemm_algo_name", &lt_gemm_algo_name);
 }

@@ -19,6 +19,7 @@ import torch.nn as nn
 
 from ..ops.cross_entropy import linear_cross_entropy
 from ..ops.flash_attn import flash_attn_xla
+from ..ops.linear import TunedLinear
 from ..ops.rmsnorm import RMSNorm, fused_add_rms_norm
 from ..ops.rope import apply_rotary_pos_emb, build_rope_cache
 from ..ops.swiglu import swiglu
@@ -92,14 +93,14 @@ class LlamaAttention(nn.Module):
         self.num_heads = h
         self.num_kv_heads = hk
         self.head_dim = cfg.hidden_size // h
-        self.q_proj = nn.Linear(cfg.hidden_size, h * self.head_dim,
-                                bias=False)
-        self.k_proj = nn.Linear(cfg.hidden_size, hk * self.head_dim,
-                                bias=False)
-        self.v_proj = nn.Linear(cfg.hidden_size, hk * self.head_dim,
-                                bias=False)
-        self.o_proj = nn.Linear(h * self.head_dim, cfg.hidden_size,
-                                bias=False)
+        self.q_proj = TunedLinear(cfg.hidden_size, h * self.head_dim,
+                                  bias=False)
+        self.k_proj = TunedLinear(cfg.hidden_size, hk * self.head_dim,
+                                  bias=False)
+        self.v_proj = TunedLinear(cfg.hidden_size, hk * self.head_dim,
+                                  bias=False)
+        self.o_proj = TunedLinear(h * self.head_dim, cfg.hidden_size,
+                                  bias=False)
 
     def forward(self, x, cos, sin):
         b, s, _ = x.shape
@@ -145,12 +146,12 @@ class LlamaMLP(nn.Module):
 
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
-        self.gate_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size,
+        self.gate_proj = TunedLinear(cfg.hidden_size,
+                                     cfg.intermediate_size, bias=False)
+        self.up_proj = TunedLinear(cfg.hidden_size, cfg.intermediate_size,
                                    bias=False)
-        self.up_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size,
-                                 bias=False)
-        self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size,
-                                   bias=False)
+        self.down_proj = TunedLinear(cfg.intermediate_size, cfg.hidden_size,
+                                     bias=False)
 
     def forward(self, x):
         return self.down_proj(swiglu(self.gate_proj(x), self.up_proj(x)))

@@ -57,8 +57,9 @@ class Qwen2Attention(nn.Module):
                                 bias=True)
         self.v_proj = nn.Linear(cfg.hidden_size, hk * self.head_dim,
                                 bias=True)
-        self.o_proj = nn.Linear(h * self.head_dim, cfg.hidden_size,
-                                bias=False)
+        from ..ops.linear import TunedLinear
+        self.o_proj = TunedLinear(h * self.head_dim, cfg.hidden_size,
+                                  bias=False)
 
     def forward(self, x, cos, sin):
         b, s, _ = x.shape

@@ -100,6 +100,20 @@ def _chunk_ce_backward(ext, logits, target, lse, scale, ignore_index):
     return (soft * scale).to(logits.dtype)
 
 
+def _ce_gemm(ext, a, b, ta, tb):
+    """Chunk GEMM on the tuned hipBLASLt path (bf16 GPU), else torch."""
+    if ext is not None and a.is_cuda and a.dtype == torch.bfloat16:
+        from .linear import _algo_for
+        m = a.shape[1] if ta else a.shape[0]
+        k = a.shape[0] if ta else a.shape[1]
+        n = b.shape[0] if tb else b.shape[1]
+        a = a if a.is_contiguous() else a.contiguous()
+        return ext.lt_gemm(a, b, ta, tb, _algo_for(m, n, k, ta, tb))
+    op_a = a.t() if ta else a
+    op_b = b.t() if tb else b
+    return (op_a @ op_b).contiguous()
+
+
 class _LinearCrossEntropy(torch.autograd.Function):
     """y = CE(x @ W^T, target) without materializing full logits.
 
@@ -121,7 +135,7 @@ class _LinearCrossEntropy(torch.autograd.Function):
         lse_all = torch.empty(N, dtype=torch.float32, device=x.device)
         for s in range(0, N, _LinearCrossEntropy.CHUNK):
             e = min(N, s + _LinearCrossEntropy.CHUNK)
-            logits = (x[s:e] @ weight.t()).contiguous()
+            logits = _ce_gemm(ext, x[s:e], weight, False, True)
             loss_sum, nvalid, lse = _chunk_ce_forward(
                 ext, logits, target[s:e], ignore_index)
             total += loss_sum.float()
@@ -144,11 +158,11 @@ class _LinearCrossEntropy(torch.autograd.Function):
         dw = torch.zeros_like(weight, dtype=torch.float32)
         for s in range(0, N, _LinearCrossEntropy.CHUNK):
             e = min(N, s + _LinearCrossEntropy.CHUNK)
-            logits = (x[s:e] @ weight.t()).contiguous()
+            logits = _ce_gemm(ext, x[s:e], weight, False, True)
             dl = _chunk_ce_backward(ext, logits, target[s:e], lse_all[s:e],
                                     scale, ignore_index)
-            dx[s:e] = dl @ weight
-            dw += (dl.t() @ x[s:e]).float()
+            dx[s:e] = _ce_gemm(ext, dl, weight, False, False)
+            dw += _ce_gemm(ext, dl, x[s:e], True, False).float()
         return dx, dw.to(weight.dtype), None, None
 
 
