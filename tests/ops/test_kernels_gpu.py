@@ -553,3 +553,48 @@ def test_fa_varlen_cross_lengths():
                          d ** -0.5, False, (-1, -1))
     err = (out.float() - ref.float()).abs().max().item()
     assert err < 2e-2, err
+
+
+@pytest.mark.parametrize("op", ["nt", "nn", "tn"])
+def test_lt_gemm_parity(op):
+    """lt_gemm (hipBLASLt explicit-algo path) vs torch.matmul for the three
+    linear-layer GEMM patterns, heuristic and first tuned candidate."""
+    from torchacc_amd.ops._backend import require_extension
+    ext = require_extension()
+    torch.manual_seed(0)
+    m, n, k = 512, 384, 256
+    ta, tb = op[0] == "t", op[1] == "t"
+    a = torch.randn((k, m) if ta else (m, k), device="cuda",
+                    dtype=torch.bfloat16)
+    b = torch.randn((n, k) if tb else (k, n), device="cuda",
+                    dtype=torch.bfloat16)
+    ref = (a.t() if ta else a).float() @ (b.t() if tb else b).float()
+    out = ext.lt_gemm(a, b, ta, tb, -1)
+    err = (out.float() - ref).abs().max() / ref.abs().max()
+    assert err.item() < 2e-2, err.item()
+    cands = ext.lt_gemm_candidates(m, n, k, ta, tb, 256)
+    assert len(cands) > 0
+    out2 = ext.lt_gemm(a, b, ta, tb, cands[0])
+    err2 = (out2.float() - ref).abs().max() / ref.abs().max()
+    assert err2.item() < 2e-2, err2.item()
+
+
+def test_tuned_linear_grad_parity():
+    """tuned_linear fwd/bwd vs F.linear (fp32 reference)."""
+    from torchacc_amd.ops.linear import tuned_linear
+    torch.manual_seed(0)
+    x = torch.randn(64, 4, 256, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = torch.randn(512, 256, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    xr = x.detach().float().requires_grad_(True)
+    wr = w.detach().float().requires_grad_(True)
+    out = tuned_linear(x, w)
+    out.sum().backward()
+    ref = torch.nn.functional.linear(xr, wr)
+    ref.sum().backward()
+    assert (out.float() - ref).abs().max() / ref.abs().max() < 2e-2
+    assert (x.grad.float() - xr.grad).abs().max() / \
+        xr.grad.abs().max().clamp_min(1) < 2e-2
+    assert (w.grad.float() - wr.grad).abs().max() / \
+        wr.grad.abs().max().clamp_min(1) < 2e-2
