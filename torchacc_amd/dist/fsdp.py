@@ -193,16 +193,23 @@ class FlatParamUnit:
         if full_grad is None:
             return
         if rs_stream is not None:
-            rs_stream.wait_stream(torch.cuda.current_stream())
+            compute = torch.cuda.current_stream()
+            # allocate the shard grad on the COMPUTE stream: it outlives this
+            # backward (consumed by the optimizer, freed by zero_grad), so
+            # its lifetime must be tracked by the stream that consumes it —
+            # allocating it on the side stream would let the caching
+            # allocator hand its block to a later reduce while the optimizer
+            # still reads it
+            if self.ws > 1:
+                out = torch.empty(self.shard_numel, dtype=full_grad.dtype,
+                                  device=full_grad.device)
+            else:
+                out = full_grad
+            rs_stream.wait_stream(compute)
             with torch.cuda.stream(rs_stream):
                 if self.ws > 1:
-                    out = torch.empty(self.shard_numel,
-                                      dtype=full_grad.dtype,
-                                      device=full_grad.device)
                     dist.reduce_scatter_tensor(out, full_grad,
                                                group=self.group)
-                else:
-                    out = full_grad
                 if dp_group is not None:
                     dist.all_reduce(out, group=dp_group)
                 if grad_scale != 1.0:
@@ -211,6 +218,7 @@ class FlatParamUnit:
                     self.shard.grad = out
                 else:
                     self.shard.grad.add_(out)
+                    out.record_stream(rs_stream)
                 full_grad.record_stream(rs_stream)
             ev = torch.cuda.Event()
             ev.record(rs_stream)
