@@ -16,7 +16,9 @@
 #include <hipblaslt/hipblaslt.h>
 #include <hipblaslt/hipblaslt-ext.hpp>
 
+#include <memory>
 #include <mutex>
+#include <string>
 #include <unordered_map>
 #include <vector>
 
@@ -137,22 +139,34 @@ std::vector<int64_t> lt_gemm_candidates(int64_t m, int64_t n, int64_t k,
   return out;
 }
 
-// out[M,N] = opA(a) @ opB(b); algo_index < 0 -> heuristic top-1.
-torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, bool ta, bool tb,
-                      int64_t algo_index, c10::optional<torch::Tensor> out_opt) {
-  int64_t m, n, k;
-  check_inputs(a, b, ta, tb, m, n, k);
-  torch::Tensor out =
-      out_opt.has_value() ? *out_opt : torch::empty({m, n}, a.options());
-  TORCH_CHECK(out.is_contiguous() && out.size(0) == m && out.size(1) == n);
-  LtDesc d;
-  build_desc(d, m, n, k, ta, tb, a.size(1), b.size(1));
+namespace {
 
-  float alpha = 1.0f, beta = 0.0f;
-  auto stream = at::hip::getCurrentHIPStream();
-  auto& ws = lt_workspace();
-
+// Per-problem plan cache: descriptors + resolved algo. The algo lookup
+// (getAlgosFromIndex / heuristic query) costs host-side milliseconds — at
+// ~460 GEMM calls per training step an uncached lookup dominates the step.
+struct LtPlan {
+  LtDesc desc;
   hipblasLtMatmulAlgo_t algo;
+};
+
+std::unordered_map<std::string, std::unique_ptr<LtPlan>> plan_cache;
+std::mutex plan_mutex;
+
+LtPlan* get_plan(int64_t m, int64_t n, int64_t k, bool ta, bool tb,
+                 int64_t algo_index) {
+  char key[96];
+  snprintf(key, sizeof(key), "%ld,%ld,%ld,%d%d,%ld", (long)m, (long)n,
+           (long)k, (int)ta, (int)tb, (long)algo_index);
+  std::lock_guard<std::mutex> lock(plan_mutex);
+  auto it = plan_cache.find(key);
+  if (it != plan_cache.end()) return it->second.get();
+
+  auto plan = std::make_unique<LtPlan>();
+  int64_t lda = ta ? m : k;
+  int64_t ldb = tb ? k : n;
+  build_desc(plan->desc, m, n, k, ta, tb, lda, ldb);
+  float alpha = 1.0f, beta = 0.0f;
+  auto& ws = lt_workspace();
   bool have_algo = false;
   if (algo_index >= 0) {
     std::vector<hipblasLtMatmulHeuristicResult_t> fetched;
@@ -161,10 +175,10 @@ torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, bool ta, bool tb,
     if (st == HIPBLAS_STATUS_SUCCESS && !fetched.empty()) {
       size_t ws_need = 0;
       auto sup = hipblaslt_ext::matmulIsAlgoSupported(
-          lt_handle(), d.op, &alpha, d.la, d.lb, &beta, d.lc, d.lc,
-          fetched[0].algo, ws_need);
+          lt_handle(), plan->desc.op, &alpha, plan->desc.la, plan->desc.lb,
+          &beta, plan->desc.lc, plan->desc.lc, fetched[0].algo, ws_need);
       if (sup == HIPBLAS_STATUS_SUCCESS && ws_need <= (size_t)ws.numel()) {
-        algo = fetched[0].algo;
+        plan->algo = fetched[0].algo;
         have_algo = true;
       }
     }
@@ -180,17 +194,36 @@ torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, bool ta, bool tb,
     hipblasLtMatmulHeuristicResult_t res;
     int returned = 0;
     HIPBLASLT_CHECK(hipblasLtMatmulAlgoGetHeuristic(
-        lt_handle(), d.op, d.la, d.lb, d.lc, d.lc, pref, 1, &res, &returned));
+        lt_handle(), plan->desc.op, plan->desc.la, plan->desc.lb,
+        plan->desc.lc, plan->desc.lc, pref, 1, &res, &returned));
     hipblasLtMatmulPreferenceDestroy(pref);
     TORCH_CHECK(returned > 0, "lt_gemm: no heuristic algo for ", m, "x", n,
                 "x", k);
-    algo = res.algo;
-    have_algo = true;
+    plan->algo = res.algo;
   }
+  LtPlan* raw = plan.get();
+  plan_cache[key] = std::move(plan);
+  return raw;
+}
 
+}  // namespace
+
+// out[M,N] = opA(a) @ opB(b); algo_index < 0 -> heuristic top-1 (cached).
+torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, bool ta, bool tb,
+                      int64_t algo_index, c10::optional<torch::Tensor> out_opt) {
+  int64_t m, n, k;
+  check_inputs(a, b, ta, tb, m, n, k);
+  torch::Tensor out =
+      out_opt.has_value() ? *out_opt : torch::empty({m, n}, a.options());
+  TORCH_CHECK(out.is_contiguous() && out.size(0) == m && out.size(1) == n);
+  LtPlan* plan = get_plan(m, n, k, ta, tb, algo_index);
+  float alpha = 1.0f, beta = 0.0f;
+  auto stream = at::hip::getCurrentHIPStream();
+  auto& ws = lt_workspace();
   HIPBLASLT_CHECK(hipblasLtMatmul(
-      lt_handle(), d.op, &alpha, b.data_ptr(), d.la, a.data_ptr(), d.lb,
-      &beta, out.data_ptr(), d.lc, out.data_ptr(), d.lc, &algo, ws.data_ptr(),
+      lt_handle(), plan->desc.op, &alpha, b.data_ptr(), plan->desc.la,
+      a.data_ptr(), plan->desc.lb, &beta, out.data_ptr(), plan->desc.lc,
+      out.data_ptr(), plan->desc.lc, &plan->algo, ws.data_ptr(),
       (size_t)ws.numel(), stream));
   return out;
 }

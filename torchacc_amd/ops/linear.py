@@ -76,7 +76,7 @@ class _LtLinear(torch.autograd.Function):
 def tuned_linear(x: torch.Tensor, weight: torch.Tensor,
                  bias: Optional[torch.Tensor] = None) -> torch.Tensor:
     """F.linear drop-in; hipBLASLt with pinned algos on GPU bf16."""
-    ext = _load()
+    ext = None if os.environ.get("TA_DISABLE_TUNED_GEMM") == "1" else _load()
     if (ext is None or not x.is_cuda or bias is not None
             or x.dtype != torch.bfloat16 or weight.dtype != torch.bfloat16):
         return F.linear(x, weight, bias)
