@@ -42,6 +42,9 @@ def parse_args():
                         "of retaining attention outputs (selective AC)")
     p.add_argument("--no-gc", action="store_true",
                    help="disable gradient checkpointing")
+    p.add_argument("--gc-cnt", type=int, default=None,
+                   help="checkpoint only the first N layers (288 GB HBM3E "
+                        "rarely needs all of them)")
     return p.parse_args()
 
 
@@ -104,6 +107,7 @@ def main():
         cfg.memory.gc = True
         cfg.memory.gc_cls = {"LlamaDecoderLayer", "Qwen2DecoderLayer"}
         cfg.memory.gc_selective_attn = not args.no_gc_selective
+        cfg.memory.gc_cnt = args.gc_cnt
 
     if on_gpu:
         torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", 0)))
@@ -195,6 +199,7 @@ def main():
                                     if cfg.memory.gc and
                                     cfg.memory.gc_selective_attn
                                     else cfg.memory.gc),
+                "gc_cnt": cfg.memory.gc_cnt,
                 "loss": float(last) if last is not None else None,
                 "peak_mem_gb": (round(
                     torch.cuda.max_memory_allocated() / 2**30, 2)
