@@ -206,33 +206,128 @@ class VocabParallelEmbedding(nn.Module):
         return reduce_from_tp_region(out, self.group)
 
 
+class _VocabParallelLinearCE(torch.autograd.Function):
+    """Fused linear + cross-entropy over a vocab-sharded lm_head: each rank
+    holds logits [N, V/tp]; the global log-sum-exp and the target logit are
+    assembled with three small [N]-vector all-reduces per chunk (Megatron
+    vocab-parallel loss semantics). Full [N, V] logits are never
+    materialized on any rank — the 70B config's biggest single activation.
+    """
+
+    CHUNK = 8192
+
+    @staticmethod
+    def forward(ctx, x, weight, target, group, ignore_index):
+        tp = dist.get_world_size(group)
+        v_local = weight.shape[0]
+        vocab_start = dist.get_rank(group) * v_local
+        N = x.shape[0]
+        total = x.new_zeros((), dtype=torch.float32)
+        nvalid = torch.zeros((), dtype=torch.long, device=x.device)
+        lse_all = torch.empty(N, dtype=torch.float32, device=x.device)
+        C = _VocabParallelLinearCE.CHUNK
+        for s in range(0, N, C):
+            e = min(N, s + C)
+            lf = (x[s:e] @ weight.t()).float()
+            gmax = lf.max(-1).values
+            if tp > 1:
+                dist.all_reduce(gmax, op=dist.ReduceOp.MAX, group=group)
+            sumexp = (lf - gmax.unsqueeze(-1)).exp().sum(-1)
+            if tp > 1:
+                dist.all_reduce(sumexp, group=group)
+            lse = gmax + sumexp.log()
+            tgt = target[s:e]
+            valid = tgt != ignore_index
+            in_part = valid & (tgt >= vocab_start) & \
+                (tgt < vocab_start + v_local)
+            tlocal = (tgt - vocab_start).masked_fill(~in_part, 0)
+            tlogit = lf.gather(1, tlocal.unsqueeze(-1)).squeeze(-1) * in_part
+            if tp > 1:
+                dist.all_reduce(tlogit, group=group)
+            total += ((lse - tlogit) * valid).sum()
+            nvalid += valid.sum()
+            lse_all[s:e] = lse
+        ctx.save_for_backward(x, weight, target, lse_all, nvalid)
+        ctx.group = group
+        ctx.ignore_index = ignore_index
+        ctx.vocab_start = vocab_start
+        return total / nvalid.clamp_min(1).float()
+
+    @staticmethod
+    def backward(ctx, dloss):
+        x, weight, target, lse_all, nvalid = ctx.saved_tensors
+        group = ctx.group
+        tp = dist.get_world_size(group)
+        v_local = weight.shape[0]
+        vocab_start = ctx.vocab_start
+        N = x.shape[0]
+        scale = dloss.float() / nvalid.clamp_min(1).float()
+        dx = torch.empty_like(x)
+        dw = torch.zeros_like(weight, dtype=torch.float32)
+        C = _VocabParallelLinearCE.CHUNK
+        for s in range(0, N, C):
+            e = min(N, s + C)
+            lf = (x[s:e] @ weight.t()).float()
+            soft = (lf - lse_all[s:e].unsqueeze(-1)).exp()
+            tgt = target[s:e]
+            valid = tgt != ctx.ignore_index
+            in_part = valid & (tgt >= vocab_start) & \
+                (tgt < vocab_start + v_local)
+            tlocal = (tgt - vocab_start).masked_fill(~in_part, 0)
+            soft.scatter_add_(
+                1, tlocal.unsqueeze(-1),
+                -in_part.unsqueeze(-1).to(soft.dtype))
+            dl = (soft * (scale * valid).unsqueeze(-1)).to(x.dtype)
+            dx[s:e] = dl @ weight
+            dw += (dl.t() @ x[s:e]).float()
+        if tp > 1:
+            # each rank's dl covers only its vocab shard: the true dx sums
+            # the partial products over the group
+            dist.all_reduce(dx, group=group)
+        return dx, dw.to(weight.dtype), None, None, None
+
+
+def vocab_parallel_linear_cross_entropy(x, weight, target, group,
+                                        ignore_index: int = -100):
+    return _VocabParallelLinearCE.apply(x, weight, target, group,
+                                        ignore_index)
+
+
 def parallelize_module(model: nn.Module, config) -> nn.Module:
-    """Shard the native Llama family over the TP group (q/k/v/gate/up
-    column, o/down row, heads divided). Non-Llama models: apply the classes
-    manually."""
+    """Shard a Llama-family model over the TP group: q/k/v/gate/up column,
+    o/down row, heads divided, embed_tokens + lm_head vocab-parallel with
+    the CE loss computed on sharded logits. Matching is duck-typed on the
+    projection attribute names, so native models AND HF transformers
+    modules are both covered (the reference expressed TP as GSPMD
+    annotations over any model, dist/tp.py:4-5; this is the eager
+    equivalent). Other models: apply the classes manually."""
     mesh = config.get_mesh()
     group = mesh.get_tp_proc_group()
     tp = mesh.get_tp_num()
     if tp == 1 or group is None:
         return model
-    from ..models.llama import LlamaAttention, LlamaMLP
-    from ..models.qwen2 import Qwen2Attention
     n_attn = n_mlp = 0
     for mod in model.modules():
-        if isinstance(mod, (LlamaAttention, Qwen2Attention)):
-            assert mod.num_heads % tp == 0, \
-                f"attention heads {mod.num_heads} not divisible by tp {tp}"
-            assert mod.num_kv_heads % tp == 0, \
-                (f"kv heads {mod.num_kv_heads} not divisible by tp {tp}; "
-                 "use a smaller tp degree for this GQA config")
+        if all(hasattr(mod, a) for a in ("q_proj", "k_proj", "v_proj",
+                                         "o_proj")) and \
+                isinstance(mod.q_proj, nn.Linear):
+            assert mod.q_proj.out_features % tp == 0, \
+                f"q_proj {mod.q_proj.out_features} not divisible by tp {tp}"
+            assert mod.k_proj.out_features % tp == 0, \
+                (f"kv projection {mod.k_proj.out_features} not divisible by "
+                 f"tp {tp}; use a smaller tp degree for this GQA config")
             mod.q_proj = ColumnParallelLinear.from_linear(mod.q_proj, group)
             mod.k_proj = ColumnParallelLinear.from_linear(mod.k_proj, group)
             mod.v_proj = ColumnParallelLinear.from_linear(mod.v_proj, group)
             mod.o_proj = RowParallelLinear.from_linear(mod.o_proj, group)
-            mod.num_heads //= tp
-            mod.num_kv_heads //= tp
+            if hasattr(mod, "num_heads"):
+                mod.num_heads //= tp
+            if hasattr(mod, "num_kv_heads"):
+                mod.num_kv_heads //= tp
             n_attn += 1
-        elif isinstance(mod, LlamaMLP):
+        elif all(hasattr(mod, a) for a in ("gate_proj", "up_proj",
+                                           "down_proj")) and \
+                isinstance(mod.gate_proj, nn.Linear):
             mod.gate_proj = ColumnParallelLinear.from_linear(
                 mod.gate_proj, group)
             mod.up_proj = ColumnParallelLinear.from_linear(mod.up_proj,
@@ -240,6 +335,7 @@ def parallelize_module(model: nn.Module, config) -> nn.Module:
             mod.down_proj = RowParallelLinear.from_linear(mod.down_proj,
                                                           group)
             n_mlp += 1
+    _parallelize_embedding_and_head(model, group, tp)
     if n_attn == 0 and n_mlp == 0:
         logger.warning(
             "parallelize_module: no Llama attention/MLP modules found; "
@@ -248,6 +344,41 @@ def parallelize_module(model: nn.Module, config) -> nn.Module:
         logger.info("TP=%d: sharded %d attention + %d MLP blocks", tp,
                     n_attn, n_mlp)
     return model
+
+
+def _parallelize_embedding_and_head(model: nn.Module, group, tp: int):
+    """Vocab-shard embed_tokens and lm_head (the 70B config's fattest
+    replicated blocks) wherever the model exposes them under the standard
+    names; the loss is computed vocab-parallel by the model forward (native
+    models) or by gathering logits (``gather_output=True`` inference)."""
+    holder = None
+    for mod in model.modules():
+        if hasattr(mod, "embed_tokens") and \
+                isinstance(getattr(mod, "embed_tokens"), nn.Embedding):
+            holder = mod
+            break
+    head_holder = None
+    for mod in model.modules():
+        if hasattr(mod, "lm_head") and \
+                isinstance(getattr(mod, "lm_head"), nn.Linear):
+            head_holder = mod
+            break
+    if holder is None and head_holder is None:
+        return
+    tied = (holder is not None and head_holder is not None and
+            holder.embed_tokens.weight is head_holder.lm_head.weight)
+    if holder is not None and \
+            holder.embed_tokens.num_embeddings % tp == 0:
+        holder.embed_tokens = VocabParallelEmbedding.from_embedding(
+            holder.embed_tokens, group)
+    if head_holder is not None and \
+            head_holder.lm_head.out_features % tp == 0:
+        head = ColumnParallelLinear.from_linear(head_holder.lm_head, group)
+        head.gather_output = True  # inference path returns full logits
+        if tied and isinstance(holder.embed_tokens, VocabParallelEmbedding):
+            head.weight = holder.embed_tokens.weight
+        head_holder.lm_head = head
+        head_holder._tp_group = group
 
 
 # ---- reference-API shims ---------------------------------------------------

@@ -223,5 +223,11 @@ class LlamaForCausalLM(nn.Module):
             # shift: predict token t+1 from position t; fused linear+CE
             hs = x[:, :-1, :].reshape(-1, x.shape[-1])
             tg = labels[:, 1:].reshape(-1)
+            if getattr(self, "_tp_group", None) is not None:
+                # lm_head is vocab-sharded (dist/tp.py surgery): CE on
+                # sharded logits, never materializing [N, vocab] anywhere
+                from ..dist.tp import vocab_parallel_linear_cross_entropy
+                return vocab_parallel_linear_cross_entropy(
+                    hs, self.lm_head.weight, tg, self._tp_group)
             return linear_cross_entropy(hs, self.lm_head.weight, tg)
         return self.lm_head(x)

@@ -137,5 +137,9 @@ class Qwen2ForCausalLM(nn.Module):
         if labels is not None:
             hs = x[:, :-1, :].reshape(-1, x.shape[-1])
             tg = labels[:, 1:].reshape(-1)
+            if getattr(self, "_tp_group", None) is not None:
+                from ..dist.tp import vocab_parallel_linear_cross_entropy
+                return vocab_parallel_linear_cross_entropy(
+                    hs, self.lm_head.weight, tg, self._tp_group)
             return linear_cross_entropy(hs, self.lm_head.weight, tg)
         return self.lm_head(x)
