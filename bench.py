@@ -107,7 +107,16 @@ def main():
         cfg.memory.gc = True
         cfg.memory.gc_cls = {"LlamaDecoderLayer", "Qwen2DecoderLayer"}
         cfg.memory.gc_selective_attn = not args.no_gc_selective
-        cfg.memory.gc_cnt = args.gc_cnt
+        gc_cnt = args.gc_cnt
+        if gc_cnt is None and args.model == "llama-2-7b" \
+                and args.mode == "fsdp" \
+                and args.batch_size * args.seq_len <= 32768:
+            # 288 GB HBM3E rarely needs every layer checkpointed: 8 of 32
+            # measures 18.9k tok/s at 193 GB vs 16.6k at 118 GB for all-32
+            # (profiles/r02); headroom stays ~95 GB and shrinks further as
+            # FSDP shards states at N>1
+            gc_cnt = 8
+        cfg.memory.gc_cnt = gc_cnt
 
     if on_gpu:
         torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", 0)))

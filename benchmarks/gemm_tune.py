@@ -68,25 +68,33 @@ def shapes_for(model: str, bs: int, seq: int, tp: int):
     return sorted(shapes)
 
 
+ROT = 4  # rotate input copies so repeats never hit warm caches
+
+
 def _mk(m, n, k, op, dev):
     ta, tb = op[0] == "t", op[1] == "t"
-    a = torch.randn((k, m) if ta else (m, k), device=dev,
-                    dtype=torch.bfloat16)
-    b = torch.randn((n, k) if tb else (k, n), device=dev,
-                    dtype=torch.bfloat16)
+    mk = lambda shape: [  # noqa: E731
+        torch.randn(shape, device=dev, dtype=torch.bfloat16)
+        for _ in range(ROT)
+    ]
+    a = mk((k, m) if ta else (m, k))
+    b = mk((n, k) if tb else (k, n))
     return a, b, ta, tb
 
 
 def _time_algo(ext, a, b, ta, tb, algo, iters, out):
-    ev0, ev1 = torch.cuda.Event(True), torch.cuda.Event(True)
-    ext.lt_gemm(a, b, ta, tb, algo, out)  # warm (also validates support)
+    """Median of per-iteration times over rotating cold inputs (a single
+    hot-loop average rewards cache-resident algos that lose in context)."""
+    ev = [torch.cuda.Event(True) for _ in range(iters + 1)]
+    ext.lt_gemm(a[0], b[0], ta, tb, algo, out)  # warm/validate
     torch.cuda.synchronize()
-    ev0.record()
-    for _ in range(iters):
-        ext.lt_gemm(a, b, ta, tb, algo, out)
-    ev1.record()
+    ev[0].record()
+    for i in range(iters):
+        ext.lt_gemm(a[(i + 1) % ROT], b[(i + 1) % ROT], ta, tb, algo, out)
+        ev[i + 1].record()
     torch.cuda.synchronize()
-    return ev0.elapsed_time(ev1) / iters
+    times = sorted(ev[i].elapsed_time(ev[i + 1]) for i in range(iters))
+    return times[len(times) // 2]
 
 
 def tune_shape(ext, m, n, k, op, dev, cap, budget_s=30.0):

@@ -40,6 +40,8 @@ def _algo_table() -> dict:
 
 
 def _algo_for(m: int, n: int, k: int, ta: bool, tb: bool) -> int:
+    if os.environ.get("TA_GEMM_HEURISTIC") == "1":
+        return -1  # A/B escape hatch: heuristic top-1 everywhere
     key = f"{m},{n},{k},{'t' if ta else 'n'}{'t' if tb else 'n'}"
     return _algo_table().get(key, -1)
 
