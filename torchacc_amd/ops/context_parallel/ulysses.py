@@ -18,12 +18,21 @@ def ulysses(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
             softmax_scale=None, causal: bool = True, dropout_p: float = 0.0,
             window_size=(-1, -1),
             rope_func: Optional[Callable] = None,
-            process_group=None, **rope_kwargs) -> torch.Tensor:
+            process_group=None, attention_mask=None,
+            **rope_kwargs) -> torch.Tensor:
+    """``attention_mask`` [b, s_full] (int, right padding over the FULL
+    sequence) enables varlen attention after the sequence gather."""
     group = process_group if process_group is not None \
         else get_intra_cp_group()
     if group is None:
         if rope_func is not None:
             q, k = rope_func(q, k, **rope_kwargs)
+        if attention_mask is not None:
+            from ..flash_attn import flash_attn_varlen_xla
+            return flash_attn_varlen_xla(
+                q, k, v, attention_mask=attention_mask, dropout_p=dropout_p,
+                softmax_scale=softmax_scale, causal=causal,
+                window_size=window_size)
         return flash_attn_xla(q, k, v, dropout_p=dropout_p,
                               softmax_scale=softmax_scale, causal=causal,
                               window_size=window_size)
@@ -40,8 +49,15 @@ def ulysses(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     if rope_func is not None:
         # RoPE applied post-a2a so positions cover the full sequence
         q, k = rope_func(q, k, **rope_kwargs)
-    out = flash_attn_xla(q, k, v, dropout_p=dropout_p,
-                         softmax_scale=softmax_scale, causal=causal,
-                         window_size=window_size)
+    if attention_mask is not None:
+        from ..flash_attn import flash_attn_varlen_xla
+        out = flash_attn_varlen_xla(
+            q, k, v, attention_mask=attention_mask, dropout_p=dropout_p,
+            softmax_scale=softmax_scale, causal=causal,
+            window_size=window_size)
+    else:
+        out = flash_attn_xla(q, k, v, dropout_p=dropout_p,
+                             softmax_scale=softmax_scale, causal=causal,
+                             window_size=window_size)
     # [b, s, h/cp, d] -> [b, s/cp, h, d]
     return diff_all_to_all(out, 1, 2, group)

@@ -146,18 +146,20 @@ def update_out_and_lse(out: Optional[torch.Tensor],
     """Merge a new attention block into the running (out, lse).
 
     out/block_out: [b, s, h, d] (any dtype); lse/block_lse: [b, h, s] fp32.
-    Stable form (reference utils.py:302-343):
-        out = out - sigmoid(block_lse - lse) * (out - block_out)
-        lse = lse - logsigmoid(lse - block_lse)
+    logaddexp form (equivalent to the reference's sigmoid/logsigmoid update,
+    utils.py:302-343, but safe when either side is -inf — a fully-masked
+    varlen block contributes lse = -inf and must be a no-op):
+        new_lse = logaddexp(lse, block_lse)
+        out     = out + exp(block_lse - new_lse) * (block_out - out)
     """
     if out is None:
         return block_out.float(), block_lse
     blk = block_out.float()
-    l = lse.transpose(-2, -1).unsqueeze(-1)        # [b,s,h,1]
-    bl = block_lse.transpose(-2, -1).unsqueeze(-1)
-    out = out - torch.sigmoid(bl - l) * (out - blk)
-    lse = lse - torch.nn.functional.logsigmoid(lse - block_lse)
-    return out, lse
+    new_lse = torch.logaddexp(lse, block_lse)
+    w = torch.exp(block_lse - new_lse)             # [b,h,s], 0 for -inf blk
+    w = torch.nan_to_num(w)                        # both -inf: dead rows
+    out = out + w.transpose(-2, -1).unsqueeze(-1) * (blk - out)
+    return out, new_lse
 
 
 # ---------------------------------------------------------------------------
