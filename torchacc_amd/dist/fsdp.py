@@ -162,11 +162,24 @@ class FlatParamUnit:
     def rebuild_views(self, track_grad: bool):
         """(Re)create the per-param views into full_flat. With grad tracking
         each call builds fresh autograd edges (a view's grad_fn buffers are
-        freed after each backward, so views cannot be reused across steps)."""
+        freed after each backward, so views cannot be reused across steps).
+
+        One torch.split per unit, NOT per-param slicing: a slice of the big
+        flat leaf back-propagates by materializing a full-unit zero tensor
+        and adding (measured 83 ms/step on Llama-7B, profiles/r02); split's
+        backward is a single cat of the per-param grads."""
         src = self.full_flat if track_grad else self.full_flat.detach()
+        if not hasattr(self, "_seg_sizes"):
+            segs = sorted({(off, n) for _m, _a, off, _s, n in self.entries})
+            sizes = [n for _off, n in segs]
+            pad = self.padded_numel - sum(sizes)
+            if pad:
+                sizes.append(pad)
+            self._seg_index = {off: i for i, (off, _n) in enumerate(segs)}
+            self._seg_sizes = sizes
+        parts = torch.split(src, self._seg_sizes)
         for mod, attr, off, shape, n in self.entries:
-            v = src[off:off + n].view(shape)
-            setattr(mod, attr, v)
+            setattr(mod, attr, parts[self._seg_index[off]].view(shape))
         self._views_valid = True
 
     def reshard(self):

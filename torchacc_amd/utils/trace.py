@@ -42,8 +42,11 @@ class _Tracer(fx.Tracer):
         super().__init__(autowrap_functions=_leaf_fns())
 
     def is_leaf_module(self, m, qualname):
+        from ..ops.linear import TunedLinear
         from ..ops.rmsnorm import RMSNorm
-        if isinstance(m, RMSNorm):
+        if isinstance(m, (RMSNorm, TunedLinear)):
+            # TunedLinear.forward branches on device/dtype (GPU hipBLASLt
+            # vs F.linear) — leaf it like nn.Linear so fx never sees that
             return True
         return super().is_leaf_module(m, qualname)
 
