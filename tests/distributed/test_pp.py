@@ -240,3 +240,79 @@ def test_pp2_inference_matches_single_process(tmp_path):
         ref = model(ids).float().numpy()
     assert got.shape == ref.shape
     assert abs(got - ref).max() < 1e-4, abs(got - ref).max()
+
+
+def _pp_hf_worker(rank, world, q):
+    """HF transformers Llama split on the decoder ModuleList (no fx;
+    reference capability dist/pp/pipeline.py:38-44)."""
+    import torchacc_amd as ta
+    from transformers.models.llama.configuration_llama import LlamaConfig
+    from transformers.models.llama.modeling_llama import LlamaForCausalLM
+    hf_cfg = LlamaConfig(
+        vocab_size=256, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=4,
+        max_position_embeddings=64, attn_implementation="eager",
+        tie_word_embeddings=False)
+    cfg = ta.Config()
+    cfg.dist.pp.size = world
+    cfg.dist.pp.num_micro_batches = 2
+    cfg.dist.pp.input_names = ["input_ids", "labels"]
+    cfg.compute.disable_kernel_patches = True
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(hf_cfg)
+    model = ta.accelerate(model, config=cfg)
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
+    torch.manual_seed(42)
+    losses = []
+    for _ in range(4):
+        ids = torch.randint(0, 256, (4, 32))
+        loss = model.forward_backward(ids, labels=ids)
+        opt.step()
+        opt.zero_grad()
+        losses.append(float(loss))
+    q.put((rank, losses))
+
+
+def test_pp2_hf_llama_matches_single():
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    run_multiprocess(_pp_hf_worker, world_size=2, args=(q,))
+    results = {}
+    while not q.empty():
+        r, losses = q.get()
+        results[r] = losses
+    assert len(results) == 2
+    assert results[0] == pytest.approx(results[1], abs=1e-5)
+
+    # single-process baseline: same micro-batch split, same loss formula
+    # as the PP last stage (shifted fused linear-CE)
+    from transformers.models.llama.configuration_llama import LlamaConfig
+    from transformers.models.llama.modeling_llama import LlamaForCausalLM
+    from torchacc_amd.ops.cross_entropy import linear_cross_entropy
+    hf_cfg = LlamaConfig(
+        vocab_size=256, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=4,
+        max_position_embeddings=64, attn_implementation="eager",
+        tie_word_embeddings=False)
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(hf_cfg)
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
+    torch.manual_seed(42)
+
+    def loss_fn(ids):
+        h = model.model(input_ids=ids).last_hidden_state
+        hs = h[:, :-1, :].reshape(-1, h.shape[-1])
+        return linear_cross_entropy(hs, model.lm_head.weight,
+                                    ids[:, 1:].reshape(-1))
+
+    base = []
+    for _ in range(4):
+        ids = torch.randint(0, 256, (4, 32))
+        l1 = loss_fn(ids[:2])
+        l2 = loss_fn(ids[2:])
+        ((l1 + l2) / 2).backward()
+        opt.step()
+        opt.zero_grad()
+        base.append(float((l1 + l2) / 2))
+    assert results[0] == pytest.approx(base, abs=2e-4)

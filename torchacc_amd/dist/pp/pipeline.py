@@ -52,21 +52,32 @@ class PipelineParallel(ParallelModule):
         self._orig_forward = model.forward
 
         import torch.fx as fx
-        if not isinstance(model, fx.GraphModule):
-            try:
-                gm = trace(model, pp_cfg.input_names)
-            except Exception as e:
-                raise RuntimeError(
-                    "pipeline parallelism requires an fx-traceable model; "
-                    f"tracing {type(model).__name__} failed ({e}). "
-                    "transformers >= 5 removed its HF fx tracer, so HF "
-                    "models cannot be pipeline-split — use the native "
-                    "model families (torchacc_amd.models) for PP, or "
-                    "FSDP/DP/CP which need no tracing") from e
-        else:
-            gm = model
-        sp = _preprocess_split_points(model, pp_cfg.split_points)
-        sr = split(gm, self.num_stages, sp)
+        from .hf_split import is_hf_splittable, split_hf_model
+        sr = None
+        if not isinstance(model, fx.GraphModule) and \
+                not pp_cfg.split_points and is_hf_splittable(model):
+            # HF causal LMs: split on the decoder-layer ModuleList directly
+            # (transformers >= 5 removed HFTracer; reference capability
+            # dist/pp/pipeline.py:38-44 restored without fx)
+            sr = split_hf_model(model, self.num_stages, pp_cfg.input_names)
+        if sr is None:
+            if not isinstance(model, fx.GraphModule):
+                try:
+                    gm = trace(model, pp_cfg.input_names)
+                except Exception as e:
+                    raise RuntimeError(
+                        "pipeline parallelism requires an fx-traceable "
+                        f"model; tracing {type(model).__name__} failed "
+                        f"({e}). transformers >= 5 removed its HF fx "
+                        "tracer; HF causal LMs are split on their decoder "
+                        "ModuleList instead — this model matches neither. "
+                        "Use the native model families (torchacc_amd."
+                        "models) for PP, or FSDP/DP/CP which need no "
+                        "tracing") from e
+            else:
+                gm = model
+            sp = _preprocess_split_points(model, pp_cfg.split_points)
+            sr = split(gm, self.num_stages, sp)
         if self.mesh.global_rank == 0:
             for i, spec in enumerate(sr.specs):
                 logger.info("stage %d: %d inputs, send %s", i,
