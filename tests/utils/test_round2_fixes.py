@@ -136,3 +136,58 @@ def test_adamw_skips_on_found_inf():
     p.grad = torch.randn(8)
     opt.step(found_inf=torch.ones(()))
     assert torch.equal(p.detach(), before)
+
+
+def test_hf_lce_forward_loss_parity_and_no_logits():
+    """HF causal-LM forward routes lm_head+loss through chunked fused
+    linear-CE (Liger lce_forward parity, reference ops/liger.py:75-76):
+    loss matches the unpatched model, logits are never materialized."""
+    pytest.importorskip("transformers")
+    from transformers.models.llama.configuration_llama import LlamaConfig
+    from transformers.models.llama.modeling_llama import LlamaForCausalLM
+
+    from torchacc_amd.utils.patch import (_ORIG_CAUSAL_FWD,
+                                          apply_fused_kernel_patches)
+
+    apply_fused_kernel_patches()
+    assert LlamaForCausalLM in _ORIG_CAUSAL_FWD
+    cfg = LlamaConfig(vocab_size=128, hidden_size=32, intermediate_size=64,
+                      num_hidden_layers=2, num_attention_heads=4,
+                      num_key_value_heads=4, max_position_embeddings=64,
+                      attn_implementation="eager")
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(cfg)
+    ids = torch.randint(0, 128, (2, 16))
+    out = model(input_ids=ids, labels=ids)
+    assert out.logits is None, "fused path must not materialize logits"
+    ref = _ORIG_CAUSAL_FWD[LlamaForCausalLM](model, input_ids=ids,
+                                             labels=ids)
+    assert abs(float(out.loss) - float(ref.loss)) < 1e-4
+    # inference path (no labels) untouched: full logits
+    infer = model(input_ids=ids)
+    assert infer.logits is not None and infer.logits.shape[-1] == 128
+    # grad-accumulation normalization contract (HF num_items_in_batch)
+    out2 = model(input_ids=ids, labels=ids, num_items_in_batch=60)
+    want = float(ref.loss) * 2 * 15 / 60
+    assert abs(float(out2.loss) - want) < 1e-4
+
+
+def test_hf_rope_patch_matches_eager():
+    pytest.importorskip("transformers")
+    from transformers.models.llama.modeling_llama import rotate_half
+
+    from torchacc_amd.utils.patch import _hf_rope_forward
+
+    torch.manual_seed(0)
+    b, h, s, d = 2, 4, 16, 32
+    q = torch.randn(b, h, s, d)
+    k = torch.randn(b, 2, s, d)
+    half = torch.randn(1, s, d // 2)
+    cos = torch.cat([half.cos(), half.cos()], dim=-1)
+    sin = torch.cat([half.sin(), half.sin()], dim=-1)
+    qo, ko = _hf_rope_forward(q, k, cos, sin)
+    c, si = cos.unsqueeze(1), sin.unsqueeze(1)
+    q_ref = q * c + rotate_half(q) * si
+    k_ref = k * c + rotate_half(k) * si
+    assert torch.allclose(qo, q_ref, atol=1e-5)
+    assert torch.allclose(ko, k_ref, atol=1e-5)

@@ -156,19 +156,89 @@ def _patched_mlp_forward(self, x):
     return self.down_proj(swiglu(self.gate_proj(x), self.up_proj(x)))
 
 
+_ORIG_CAUSAL_FWD = {}
+
+
+def _lce_causal_forward(self, input_ids=None, attention_mask=None,
+                        position_ids=None, past_key_values=None,
+                        inputs_embeds=None, labels=None, use_cache=None,
+                        output_attentions=None, output_hidden_states=None,
+                        return_dict=None, cache_position=None,
+                        logits_to_keep=0, **kwargs):
+    """HF causal-LM forward with the loss fused into the lm_head GEMM —
+    the full [b, s, vocab] logits tensor is never materialized (Liger
+    lce_forward parity, reference ops/liger.py:75-76). ``logits`` in the
+    returned output is None when labels are given, exactly like Liger."""
+    orig = _ORIG_CAUSAL_FWD[type(self)]
+    num_items = kwargs.pop("num_items_in_batch", None)
+    if labels is None or output_attentions or output_hidden_states or \
+            logits_to_keep not in (0, None):
+        return orig(self, input_ids=input_ids,
+                    attention_mask=attention_mask,
+                    position_ids=position_ids,
+                    past_key_values=past_key_values,
+                    inputs_embeds=inputs_embeds, labels=labels,
+                    use_cache=use_cache,
+                    output_attentions=output_attentions,
+                    output_hidden_states=output_hidden_states,
+                    return_dict=return_dict, cache_position=cache_position,
+                    logits_to_keep=logits_to_keep or 0, **kwargs)
+    outputs = self.model(
+        input_ids=input_ids, attention_mask=attention_mask,
+        position_ids=position_ids, past_key_values=past_key_values,
+        inputs_embeds=inputs_embeds, use_cache=use_cache,
+        cache_position=cache_position, **kwargs)
+    hidden = getattr(outputs, "last_hidden_state", None)
+    if hidden is None:
+        hidden = outputs[0]
+    hs = hidden[:, :-1, :].reshape(-1, hidden.shape[-1])
+    tg = labels[:, 1:].reshape(-1)
+    from ..ops.cross_entropy import linear_cross_entropy
+    loss = linear_cross_entropy(hs, self.lm_head.weight, tg)
+    if num_items is not None:
+        # HF Trainer gradient-accumulation normalization: divide the token
+        # SUM by the global token count instead of the local mean
+        nvalid = (tg != -100).sum().clamp_min(1)
+        loss = loss * nvalid.to(loss.dtype) / num_items
+    from transformers.modeling_outputs import CausalLMOutputWithPast
+    return CausalLMOutputWithPast(
+        loss=loss, logits=None,
+        past_key_values=getattr(outputs, "past_key_values", None),
+        hidden_states=getattr(outputs, "hidden_states", None),
+        attentions=None)
+
+
+def _hf_rope_forward(q, k, cos, sin, position_ids=None, unsqueeze_dim=1):
+    """HF apply_rotary_pos_emb signature (q/k [b, h, s, d]; cos/sin
+    [b, s, d] duplicated halves) over the fused CDNA4 RoPE kernel."""
+    import torch.nn.functional  # noqa: F401 (parity with HF imports)
+    if cos.dim() != 3 or cos.shape[0] != 1 or unsqueeze_dim != 1:
+        # per-batch position tables (packing): keep HF's eager math
+        from transformers.models.llama.modeling_llama import rotate_half
+        c = cos.unsqueeze(unsqueeze_dim)
+        s = sin.unsqueeze(unsqueeze_dim)
+        return q * c + rotate_half(q) * s, k * c + rotate_half(k) * s
+    d2 = q.shape[-1] // 2
+    from ..ops.rope import _RoPE
+    qo, ko = _RoPE.apply(q.permute(0, 2, 1, 3), k.permute(0, 2, 1, 3),
+                         cos[0, :, :d2], sin[0, :, :d2])
+    return qo.permute(0, 2, 1, 3), ko.permute(0, 2, 1, 3)
+
+
 def apply_fused_kernel_patches(model: Optional[torch.nn.Module] = None
                                ) -> bool:
-    """Swap HF Llama/Qwen2 RMSNorm and MLP forwards for the fused CDNA4
-    kernels (Liger-equivalent, reference ops/liger.py:133)."""
+    """Swap HF Llama/Qwen2 RMSNorm / SwiGLU-MLP / RoPE forwards for the
+    fused CDNA4 kernels and fuse lm_head+loss into chunked
+    linear-cross-entropy (Liger-equivalent, reference ops/liger.py:133)."""
     tf = _transformers()
     if tf is None:
         return False
     patched = False
-    for mod_name, rms_name, mlp_name in (
+    for mod_name, rms_name, mlp_name, lm_name in (
             ("transformers.models.llama.modeling_llama", "LlamaRMSNorm",
-             "LlamaMLP"),
+             "LlamaMLP", "LlamaForCausalLM"),
             ("transformers.models.qwen2.modeling_qwen2", "Qwen2RMSNorm",
-             "Qwen2MLP")):
+             "Qwen2MLP", "Qwen2ForCausalLM")):
         try:
             import importlib
             m = importlib.import_module(mod_name)
@@ -182,8 +252,15 @@ def apply_fused_kernel_patches(model: Optional[torch.nn.Module] = None
         if mlp is not None:
             mlp.forward = _patched_mlp_forward
             patched = True
+        lm = getattr(m, lm_name, None)
+        if lm is not None and lm not in _ORIG_CAUSAL_FWD:
+            _ORIG_CAUSAL_FWD[lm] = lm.forward
+            lm.forward = _lce_causal_forward
+        if hasattr(m, "apply_rotary_pos_emb"):
+            m.apply_rotary_pos_emb = _hf_rope_forward
     if patched:
-        logger.info("applied fused RMSNorm/SwiGLU patches to HF models")
+        logger.info("applied fused RMSNorm/SwiGLU/RoPE/linear-CE patches "
+                    "to HF models")
     return patched
 
 
