@@ -126,3 +126,86 @@ def test_ckpt_cli_entry(tmp_path):
     assert os.path.exists(os.path.join(out_dir, "consolidated_model.pth"))
     assert len([f for f in os.listdir(out_dir)
                 if f.startswith("rank-") and f.endswith("-model.pth")]) == 2
+
+
+def _interop_worker(rank, world, out_dir):
+    """FSDP-train -> full optimizer state -> PLAIN torch optimizer resumes
+    identically (reference interop, dist/fsdp.py:291-424)."""
+    import torchacc_amd as ta
+    from torchacc_amd.dist.state_dict_utils import (full_optim_state_dict,
+                                                    to_torch_optim_state_dict)
+    cfg = ta.Config()
+    cfg.dist.fsdp.size = world
+    cfg.dist.fsdp.wrap_layer_cls = {"LlamaDecoderLayer"}
+    model = _make_model()
+    wrapped = ta.accelerate(model, config=cfg)
+    fsdp = wrapped.fsdp_wrapper
+    # SGD+momentum: linear in the gradients, so the plain-optimizer replay
+    # must match to fp32 noise (AdamW's normalized update amplifies
+    # last-bit grad differences by +-lr near zero-curvature params)
+    opt = torch.optim.SGD(wrapped.parameters(), lr=1e-2, momentum=0.9)
+    torch.manual_seed(7)
+    data = [torch.randint(0, 1024, (2, 32)) for _ in range(5)]
+    for i in range(3):
+        loss = wrapped(data[i], labels=data[i])
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+    full_w = wrapped.full_state_dict()
+    full_o = full_optim_state_dict(fsdp, opt)
+    # continue the FSDP run two more steps -> target losses
+    cont = []
+    for i in range(3, 5):
+        loss = wrapped(data[i], labels=data[i])
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        cont.append(float(loss))
+    if rank == 0:
+        plain = _make_model(seed=123)  # different init: weights come from ckpt
+        missing, unexpected = plain.load_state_dict(full_w, strict=False)
+        assert not [m for m in missing if not m.startswith("rope_")], missing
+        popt = torch.optim.SGD(plain.parameters(), lr=1e-2, momentum=0.9)
+        popt.load_state_dict(to_torch_optim_state_dict(full_o, plain))
+        replay = []
+        for i in range(3, 5):
+            loss = plain(data[i], labels=data[i])
+            loss.backward()
+            popt.step()
+            popt.zero_grad()
+            replay.append(float(loss))
+        import json
+        with open(f"{out_dir}/interop.json", "w") as f:
+            json.dump({"cont": cont, "replay": replay}, f)
+
+
+def test_full_optim_state_plain_torch_interop(tmp_path):
+    import json
+    run_multiprocess(_interop_worker, world_size=2, args=(str(tmp_path),))
+    with open(tmp_path / "interop.json") as f:
+        r = json.load(f)
+    for a, b in zip(r["cont"], r["replay"]):
+        assert abs(a - b) < 2e-4, (r["cont"], r["replay"])
+
+
+def _clip_inf_worker(rank, world, out_dir):
+    import torchacc_amd as ta
+    cfg = ta.Config()
+    cfg.dist.fsdp.size = world
+    cfg.dist.fsdp.wrap_layer_cls = {"LlamaDecoderLayer"}
+    model = _make_model()
+    ref = _make_model()
+    wrapped = ta.accelerate(model, config=cfg)
+    torch.manual_seed(9)
+    ids = torch.randint(0, 1024, (2, 32))
+    wrapped(ids, labels=ids).backward()
+    got_inf = float(wrapped.clip_grad_norm_(1e-3, norm_type=float("inf")))
+    ref(ids, labels=ids).backward()
+    want_inf = float(torch.nn.utils.clip_grad_norm_(
+        ref.parameters(), 1e-3, norm_type=float("inf")))
+    assert abs(got_inf - want_inf) / max(want_inf, 1e-9) < 1e-3, \
+        (got_inf, want_inf)
+
+
+def test_fsdp_clip_grad_norm_inf(tmp_path):
+    run_multiprocess(_clip_inf_worker, world_size=2, args=(str(tmp_path),))

@@ -64,13 +64,15 @@ class DistributedParallel(ParallelModule):
         else:
             self.module = model
 
-    def clip_grad_norm_(self, max_grad_norm: float):
+    def clip_grad_norm_(self, max_grad_norm: float, norm_type: float = 2.0):
         if self.fsdp_wrapper is not None:
-            return self.fsdp_wrapper.clip_grad_norm_(max_grad_norm)
+            return self.fsdp_wrapper.clip_grad_norm_(max_grad_norm,
+                                                     norm_type)
         if self.dp_wrapper is not None:
             return self.dp_wrapper.clip_grad_norm_(max_grad_norm)
         return torch.nn.utils.clip_grad_norm_(self.parameters(),
-                                              max_grad_norm)
+                                              max_grad_norm,
+                                              norm_type=norm_type)
 
     # FSDP optim-state passthroughs (reference distributed_parallel.py:63-111)
     def no_sync(self):

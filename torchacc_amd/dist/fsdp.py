@@ -525,14 +525,26 @@ class FullyShardedDataParallel(ParallelModule):
 
     def clip_grad_norm_(self, max_norm: float, norm_type: float = 2.0):
         """Global grad-norm over ALL shards (each rank holds 1/ws of every
-        grad, so sum-of-squares all-reduce over the fsdp group is exact)."""
+        grad, so a p-th-power sum — or max for inf — all-reduced over the
+        fsdp group is exact). Supports any p-norm plus inf, matching
+        torch.nn.utils.clip_grad_norm_ semantics."""
         grads = [u.shard.grad for u in self.units if u.shard.grad is not None]
         if not grads:
             return torch.tensor(0.0, device=self.device)
-        local = torch.stack([g.float().pow(2).sum() for g in grads]).sum()
-        if self.ws > 1:
-            dist.all_reduce(local, group=self.group)
-        total_norm = local.sqrt()
+        norm_type = float(norm_type)
+        if norm_type == float("inf"):
+            local = torch.stack([g.float().abs().max() for g in grads]).max()
+            if self.ws > 1:
+                dist.all_reduce(local, op=dist.ReduceOp.MAX,
+                                group=self.group)
+            total_norm = local
+        else:
+            local = torch.stack([
+                g.float().abs().pow(norm_type).sum() for g in grads
+            ]).sum()
+            if self.ws > 1:
+                dist.all_reduce(local, group=self.group)
+            total_norm = local.pow(1.0 / norm_type)
         clip = max_norm / (total_norm + 1e-6)
         if clip < 1:
             for g in grads:

@@ -174,6 +174,48 @@ def full_optim_state_dict(fsdp: FullyShardedDataParallel, optimizer) -> dict:
     return {"state": full_state, "param_groups": groups}
 
 
+def to_torch_optim_state_dict(full_state: dict,
+                              model: torch.nn.Module) -> dict:
+    """Convert the name-keyed full optimizer state into torch's index-keyed
+    ``Optimizer.state_dict()`` format, loadable into a PLAIN (non-FSDP)
+    optimizer built over ``model.parameters()`` — the reference's
+    unflatten-per-layer interop (reference dist/fsdp.py:291-424). The fused
+    AdamW's fp32 ``master`` entries are dropped (a plain optimizer rebuilds
+    them from the loaded weights)."""
+    name_to_idx = {n: i for i, (n, _p) in
+                   enumerate(model.named_parameters())}
+    state = {}
+    for name, entry in full_state["state"].items():
+        if name not in name_to_idx:
+            continue
+        state[name_to_idx[name]] = {
+            k: v for k, v in entry.items() if k != "master"
+        }
+    groups = []
+    for g in full_state["param_groups"]:
+        gg = dict(g)
+        gg["params"] = sorted(name_to_idx.values())
+        groups.append(gg)
+    return {"state": state, "param_groups": groups}
+
+
+def from_torch_optim_state_dict(torch_sd: dict,
+                                model: torch.nn.Module) -> dict:
+    """Inverse of :func:`to_torch_optim_state_dict`: index-keyed plain
+    optimizer state -> the name-keyed format consumed by
+    :func:`optim_state_dict_to_load`."""
+    idx_to_name = {i: n for i, (n, _p) in
+                   enumerate(model.named_parameters())}
+    state = {}
+    for idx, entry in torch_sd["state"].items():
+        name = idx_to_name.get(int(idx))
+        if name is not None:
+            state[name] = dict(entry)
+    groups = [{k: v for k, v in g.items() if k != "params"}
+              for g in torch_sd["param_groups"]]
+    return {"state": state, "param_groups": groups}
+
+
 def optim_state_dict_to_load(fsdp: FullyShardedDataParallel, full_state: dict,
                              optimizer=None) -> dict:
     """Reshard a full optimizer state dict for this rank's shards; if
