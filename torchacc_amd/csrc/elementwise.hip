@@ -103,6 +103,142 @@ __global__ void rmsnorm_bwd_kernel(const short* __restrict__ dy,
     atomicAdd(&dw[i], dw_acc[i]);
 }
 
+// Fast path for H == ITERS*2048 (block 256, 8 elems/thread/iter): the
+// add+square pass keeps its values in registers so the normalize pass
+// never re-reads the row, and dw-style reductions stay in registers.
+// Covers Llama H=4096/8192 (ITERS 2/4); other widths take the generic
+// kernels below.
+template <int ITERS>
+__global__ void add_rmsnorm_fwd_fast(const short* __restrict__ x,
+                                     const short* __restrict__ resid_in,
+                                     const short* __restrict__ w,
+                                     short* __restrict__ y,
+                                     short* __restrict__ resid_out,
+                                     float* __restrict__ inv_rms, int rows,
+                                     float eps, bool has_resid) {
+  const int H = ITERS * 2048;
+  __shared__ float red[16];
+  s16x8 wv[ITERS];
+#pragma unroll
+  for (int t = 0; t < ITERS; ++t)
+    wv[t] = *reinterpret_cast<const s16x8*>(
+        w + threadIdx.x * 8 + t * 2048);
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const short* xr = x + (long)row * H;
+    const short* rr = resid_in + (long)row * H;
+    short* yr = y + (long)row * H;
+    short* ro = resid_out + (long)row * H;
+    float ss = 0.f;
+    f32x4 rc[ITERS][2];  // the summed row, fp32, stays in registers
+#pragma unroll
+    for (int t = 0; t < ITERS; ++t) {
+      const int i = threadIdx.x * 8 + t * 2048;
+      s16x8 v = *reinterpret_cast<const s16x8*>(xr + i);
+      s16x8 o;
+      if (has_resid) {
+        s16x8 rv = *reinterpret_cast<const s16x8*>(rr + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = bf16_to_f32(v[j]) + bf16_to_f32(rv[j]);
+          // match the generic kernel: resid_out stores the bf16 ROUNDING
+          // of the sum, and the norm pass reads that rounded value
+          o[j] = f32_to_bf16(f);
+          float fr = bf16_to_f32(o[j]);
+          rc[t][j / 4][j % 4] = fr;
+          ss += fr * fr;
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = bf16_to_f32(v[j]);
+          o[j] = v[j];
+          rc[t][j / 4][j % 4] = f;
+          ss += f * f;
+        }
+      }
+      *reinterpret_cast<s16x8*>(ro + i) = o;
+    }
+    ss = block_reduce_sum<4>(ss, red);
+    float r = rsqrtf(ss / H + eps);
+    if (threadIdx.x == 0) inv_rms[row] = r;
+#pragma unroll
+    for (int t = 0; t < ITERS; ++t) {
+      const int i = threadIdx.x * 8 + t * 2048;
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = f32_to_bf16(rc[t][j / 4][j % 4] * r * bf16_to_f32(wv[t][j]));
+      *reinterpret_cast<s16x8*>(yr + i) = o;
+    }
+  }
+}
+
+template <int ITERS>
+__global__ void add_rmsnorm_bwd_fast(const short* __restrict__ dy,
+                                     const short* __restrict__ dresid,
+                                     const short* __restrict__ r_saved,
+                                     const short* __restrict__ w,
+                                     const float* __restrict__ inv_rms,
+                                     short* __restrict__ dx,
+                                     float* __restrict__ dw, int rows,
+                                     bool has_dresid) {
+  const int H = ITERS * 2048;
+  __shared__ float red[16];
+  s16x8 wv[ITERS];
+  f32x4 dwacc[ITERS][2];
+#pragma unroll
+  for (int t = 0; t < ITERS; ++t) {
+    wv[t] = *reinterpret_cast<const s16x8*>(
+        w + threadIdx.x * 8 + t * 2048);
+    dwacc[t][0] = f32x4{0.f, 0.f, 0.f, 0.f};
+    dwacc[t][1] = f32x4{0.f, 0.f, 0.f, 0.f};
+  }
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const short* xr = r_saved + (long)row * H;
+    const short* dyr = dy + (long)row * H;
+    const short* drr = dresid + (long)row * H;
+    short* dxr = dx + (long)row * H;
+    const float rinv = inv_rms[row];
+    float dot = 0.f;
+    s16x8 xc[ITERS], dc[ITERS];
+#pragma unroll
+    for (int t = 0; t < ITERS; ++t) {
+      const int i = threadIdx.x * 8 + t * 2048;
+      xc[t] = *reinterpret_cast<const s16x8*>(xr + i);
+      dc[t] = *reinterpret_cast<const s16x8*>(dyr + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        dot += bf16_to_f32(wv[t][j]) * bf16_to_f32(dc[t][j]) *
+               (bf16_to_f32(xc[t][j]) * rinv);
+    }
+    dot = block_reduce_sum<4>(dot, red) / H;
+#pragma unroll
+    for (int t = 0; t < ITERS; ++t) {
+      const int i = threadIdx.x * 8 + t * 2048;
+      s16x8 o;
+      s16x8 drv;
+      if (has_dresid) drv = *reinterpret_cast<const s16x8*>(drr + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xh = bf16_to_f32(xc[t][j]) * rinv;
+        float dyf = bf16_to_f32(dc[t][j]);
+        float g = rinv * (bf16_to_f32(wv[t][j]) * dyf - xh * dot);
+        if (has_dresid) g += bf16_to_f32(drv[j]);
+        o[j] = f32_to_bf16(g);
+        dwacc[t][j / 4][j % 4] += dyf * xh;
+      }
+      *reinterpret_cast<s16x8*>(dxr + i) = o;
+    }
+  }
+#pragma unroll
+  for (int t = 0; t < ITERS; ++t) {
+    const int i = threadIdx.x * 8 + t * 2048;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      atomicAdd(&dw[i + j], dwacc[t][j / 4][j % 4]);
+  }
+}
+
 // fused residual add + RMSNorm: r = x + resid_in; y = rmsnorm(r) * w.
 // Saves one full read+write of the residual stream per call vs separate
 // add and norm kernels (and the separate add's backward elementwise).
@@ -282,6 +418,71 @@ __global__ void rope_kernel(const short* __restrict__ x,
 // SwiGLU
 // ---------------------------------------------------------------------------
 
+
+// 2x-wide variants: two s16x8 (32 B) per thread per trip doubles the
+// loads in flight per wave — the 1x versions measured ~60% of HBM peak.
+__global__ void swiglu_fwd_kernel_v2(const short* __restrict__ g,
+                                     const short* __restrict__ u,
+                                     short* __restrict__ y, long n16) {
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; idx < n16; idx += (long)gridDim.x * blockDim.x) {
+    const long base = idx * 16;
+    s16x8 gv0 = *reinterpret_cast<const s16x8*>(g + base);
+    s16x8 gv1 = *reinterpret_cast<const s16x8*>(g + base + 8);
+    s16x8 uv0 = *reinterpret_cast<const s16x8*>(u + base);
+    s16x8 uv1 = *reinterpret_cast<const s16x8*>(u + base + 8);
+    s16x8 o0, o1;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf0 = bf16_to_f32(gv0[j]);
+      float gf1 = bf16_to_f32(gv1[j]);
+      float s0 = 1.f / (1.f + __expf(-gf0));
+      float s1 = 1.f / (1.f + __expf(-gf1));
+      o0[j] = f32_to_bf16(gf0 * s0 * bf16_to_f32(uv0[j]));
+      o1[j] = f32_to_bf16(gf1 * s1 * bf16_to_f32(uv1[j]));
+    }
+    *reinterpret_cast<s16x8*>(y + base) = o0;
+    *reinterpret_cast<s16x8*>(y + base + 8) = o1;
+  }
+}
+
+__global__ void swiglu_bwd_kernel_v2(const short* __restrict__ dy,
+                                     const short* __restrict__ g,
+                                     const short* __restrict__ u,
+                                     short* __restrict__ dg,
+                                     short* __restrict__ du, long n16) {
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; idx < n16; idx += (long)gridDim.x * blockDim.x) {
+    const long base = idx * 16;
+    s16x8 dv0 = *reinterpret_cast<const s16x8*>(dy + base);
+    s16x8 dv1 = *reinterpret_cast<const s16x8*>(dy + base + 8);
+    s16x8 gv0 = *reinterpret_cast<const s16x8*>(g + base);
+    s16x8 gv1 = *reinterpret_cast<const s16x8*>(g + base + 8);
+    s16x8 uv0 = *reinterpret_cast<const s16x8*>(u + base);
+    s16x8 uv1 = *reinterpret_cast<const s16x8*>(u + base + 8);
+    s16x8 og0, og1, ou0, ou1;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf0 = bf16_to_f32(gv0[j]);
+      float gf1 = bf16_to_f32(gv1[j]);
+      float d0 = bf16_to_f32(dv0[j]);
+      float d1 = bf16_to_f32(dv1[j]);
+      float s0 = 1.f / (1.f + __expf(-gf0));
+      float s1 = 1.f / (1.f + __expf(-gf1));
+      og0[j] = f32_to_bf16(d0 * bf16_to_f32(uv0[j]) *
+                           (s0 * (1.f + gf0 * (1.f - s0))));
+      og1[j] = f32_to_bf16(d1 * bf16_to_f32(uv1[j]) *
+                           (s1 * (1.f + gf1 * (1.f - s1))));
+      ou0[j] = f32_to_bf16(d0 * gf0 * s0);
+      ou1[j] = f32_to_bf16(d1 * gf1 * s1);
+    }
+    *reinterpret_cast<s16x8*>(dg + base) = og0;
+    *reinterpret_cast<s16x8*>(dg + base + 8) = og1;
+    *reinterpret_cast<s16x8*>(du + base) = ou0;
+    *reinterpret_cast<s16x8*>(du + base + 8) = ou1;
+  }
+}
+
 __global__ void swiglu_fwd_kernel(const short* __restrict__ g,
                                   const short* __restrict__ u,
                                   short* __restrict__ y, long n8) {
@@ -424,9 +625,18 @@ torch::Tensor swiglu_forward(torch::Tensor g, torch::Tensor u) {
   auto y = torch::empty_like(g);
   auto stream = at::hip::getCurrentHIPStream();
   const long n8 = g.numel() / 8;
-  hipLaunchKernelGGL(swiglu_fwd_kernel, dim3(grid_for(n8, 256)), dim3(256),
-                     0, stream, (const short*)g.data_ptr(),
-                     (const short*)u.data_ptr(), (short*)y.data_ptr(), n8);
+  if (g.numel() % 16 == 0) {
+    const long n16 = g.numel() / 16;
+    hipLaunchKernelGGL(swiglu_fwd_kernel_v2, dim3(grid_for(n16, 256)),
+                       dim3(256), 0, stream, (const short*)g.data_ptr(),
+                       (const short*)u.data_ptr(), (short*)y.data_ptr(),
+                       n16);
+  } else {
+    hipLaunchKernelGGL(swiglu_fwd_kernel, dim3(grid_for(n8, 256)),
+                       dim3(256), 0, stream, (const short*)g.data_ptr(),
+                       (const short*)u.data_ptr(), (short*)y.data_ptr(),
+                       n8);
+  }
   HIP_CHECK_LAST();
   return y;
 }
@@ -438,10 +648,20 @@ std::vector<torch::Tensor> swiglu_backward(torch::Tensor dy, torch::Tensor g,
   auto du = torch::empty_like(u);
   auto stream = at::hip::getCurrentHIPStream();
   const long n8 = g.numel() / 8;
-  hipLaunchKernelGGL(swiglu_bwd_kernel, dim3(grid_for(n8, 256)), dim3(256),
-                     0, stream, (const short*)dy.data_ptr(),
-                     (const short*)g.data_ptr(), (const short*)u.data_ptr(),
-                     (short*)dg.data_ptr(), (short*)du.data_ptr(), n8);
+  if (dy.numel() % 16 == 0) {
+    const long n16 = dy.numel() / 16;
+    hipLaunchKernelGGL(swiglu_bwd_kernel_v2, dim3(grid_for(n16, 256)),
+                       dim3(256), 0, stream, (const short*)dy.data_ptr(),
+                       (const short*)g.data_ptr(),
+                       (const short*)u.data_ptr(), (short*)dg.data_ptr(),
+                       (short*)du.data_ptr(), n16);
+  } else {
+    hipLaunchKernelGGL(swiglu_bwd_kernel, dim3(grid_for(n8, 256)),
+                       dim3(256), 0, stream, (const short*)dy.data_ptr(),
+                       (const short*)g.data_ptr(),
+                       (const short*)u.data_ptr(), (short*)dg.data_ptr(),
+                       (short*)du.data_ptr(), n8);
+  }
   HIP_CHECK_LAST();
   return {dg, du};
 }
@@ -461,14 +681,30 @@ std::vector<torch::Tensor> add_rmsnorm_forward(torch::Tensor x,
   auto resid_out = torch::empty_like(x);
   auto inv_rms = torch::empty({rows}, x.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(add_rmsnorm_fwd_kernel,
-                     dim3(std::min<long>(rows, 2048)), dim3(256), 0, stream,
-                     (const short*)x.data_ptr(),
-                     has_resid ? (const short*)resid.data_ptr()
-                               : (const short*)x.data_ptr(),
-                     (const short*)w.data_ptr(), (short*)y.data_ptr(),
-                     (short*)resid_out.data_ptr(), inv_rms.data_ptr<float>(),
-                     (int)rows, H, (float)eps, has_resid);
+  const short* xin = (const short*)x.data_ptr();
+  const short* rin = has_resid ? (const short*)resid.data_ptr() : xin;
+  const short* wp = (const short*)w.data_ptr();
+  short* yp = (short*)y.data_ptr();
+  short* rop = (short*)resid_out.data_ptr();
+  float* irp = inv_rms.data_ptr<float>();
+  dim3 grid(std::min<long>(rows, 2048)), block(256);
+  if (H == 4096) {
+    hipLaunchKernelGGL(add_rmsnorm_fwd_fast<2>, grid, block, 0, stream,
+                       xin, rin, wp, yp, rop, irp, (int)rows, (float)eps,
+                       has_resid);
+  } else if (H == 8192) {
+    hipLaunchKernelGGL(add_rmsnorm_fwd_fast<4>, grid, block, 0, stream,
+                       xin, rin, wp, yp, rop, irp, (int)rows, (float)eps,
+                       has_resid);
+  } else if (H == 2048) {
+    hipLaunchKernelGGL(add_rmsnorm_fwd_fast<1>, grid, block, 0, stream,
+                       xin, rin, wp, yp, rop, irp, (int)rows, (float)eps,
+                       has_resid);
+  } else {
+    hipLaunchKernelGGL(add_rmsnorm_fwd_kernel, grid, block, 0, stream,
+                       xin, rin, wp, yp, rop, irp, (int)rows, H,
+                       (float)eps, has_resid);
+  }
   HIP_CHECK_LAST();
   return {y, resid_out, inv_rms};
 }
@@ -484,16 +720,36 @@ std::vector<torch::Tensor> add_rmsnorm_backward(torch::Tensor dy,
   auto dx = torch::empty_like(r_saved);
   auto dw32 = torch::zeros({H}, r_saved.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
-  const int lds = (H + 16) * sizeof(float);
-  hipLaunchKernelGGL(add_rmsnorm_bwd_kernel,
-                     dim3(std::min<long>(rows, 512)), dim3(256), lds, stream,
-                     (const short*)dy.data_ptr(),
-                     has_dresid ? (const short*)dresid.data_ptr()
-                                : (const short*)dy.data_ptr(),
-                     (const short*)r_saved.data_ptr(),
-                     (const short*)w.data_ptr(), inv_rms.data_ptr<float>(),
-                     (short*)dx.data_ptr(), dw32.data_ptr<float>(),
-                     (int)rows, H, has_dresid);
+  const short* dyp = (const short*)dy.data_ptr();
+  const short* drp = has_dresid ? (const short*)dresid.data_ptr() : dyp;
+  const short* xp = (const short*)r_saved.data_ptr();
+  const short* wp = (const short*)w.data_ptr();
+  const float* irp = inv_rms.data_ptr<float>();
+  short* dxp = (short*)dx.data_ptr();
+  float* dwp = dw32.data_ptr<float>();
+  dim3 block(256);
+  if (H == 4096) {
+    hipLaunchKernelGGL(add_rmsnorm_bwd_fast<2>,
+                       dim3(std::min<long>(rows, 2048)), block, 0, stream,
+                       dyp, drp, xp, wp, irp, dxp, dwp, (int)rows,
+                       has_dresid);
+  } else if (H == 8192) {
+    hipLaunchKernelGGL(add_rmsnorm_bwd_fast<4>,
+                       dim3(std::min<long>(rows, 2048)), block, 0, stream,
+                       dyp, drp, xp, wp, irp, dxp, dwp, (int)rows,
+                       has_dresid);
+  } else if (H == 2048) {
+    hipLaunchKernelGGL(add_rmsnorm_bwd_fast<1>,
+                       dim3(std::min<long>(rows, 2048)), block, 0, stream,
+                       dyp, drp, xp, wp, irp, dxp, dwp, (int)rows,
+                       has_dresid);
+  } else {
+    const int lds = (H + 16) * sizeof(float);
+    hipLaunchKernelGGL(add_rmsnorm_bwd_kernel,
+                       dim3(std::min<long>(rows, 512)), block, lds, stream,
+                       dyp, drp, xp, wp, irp, dxp, dwp, (int)rows, H,
+                       has_dresid);
+  }
   HIP_CHECK_LAST();
   return {dx, dw32.to(w.scalar_type())};
 }
