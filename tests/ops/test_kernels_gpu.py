@@ -598,3 +598,91 @@ def test_tuned_linear_grad_parity():
         xr.grad.abs().max().clamp_min(1) < 2e-2
     assert (w.grad.float() - wr.grad).abs().max() / \
         wr.grad.abs().max().clamp_min(1) < 2e-2
+
+
+def test_plain_torch_scaler_is_syncfree():
+    """Through PLAIN torch.amp.GradScaler (not ours), the fused AdamW's
+    _step_supports_amp_scaling contract keeps step()+update() free of
+    host syncs (reference torch_xla syncfree semantics,
+    utils/patch.py:55-57)."""
+    from torchacc_amd.ops.adamw import AdamW
+    lin = torch.nn.Linear(64, 64).to("cuda", torch.float16)
+    opt = AdamW(lin.parameters(), lr=1e-3)
+    scaler = torch.amp.GradScaler("cuda")
+    x = torch.randn(8, 64, device="cuda", dtype=torch.float16)
+    loss = lin(x).float().pow(2).mean()
+    scaler.scale(loss).backward()
+    torch.cuda.synchronize()
+    torch.cuda.set_sync_debug_mode(2)  # error on any host sync
+    try:
+        scaler.step(opt)
+        scaler.update()
+    finally:
+        torch.cuda.set_sync_debug_mode(0)
+    torch.cuda.synchronize()
+    assert torch.isfinite(lin.weight.float()).all()
+
+
+def test_fp16_flash_attn_casts_to_bf16_kernels():
+    """fp16 q/k/v run the bf16 MFMA kernels (explicit cast), never the
+    fp32 composite (reference accepts fp16, ops/flash_attn.py:324-325)."""
+    from torchacc_amd.ops.flash_attn import flash_attn_xla
+    torch.manual_seed(0)
+    b, s, h, d = 2, 256, 4, 128
+    q = torch.randn(b, s, h, d, device="cuda", dtype=torch.float16,
+                    requires_grad=True)
+    k = torch.randn(b, s, h, d, device="cuda", dtype=torch.float16,
+                    requires_grad=True)
+    v = torch.randn(b, s, h, d, device="cuda", dtype=torch.float16,
+                    requires_grad=True)
+    before = torch.cuda.max_memory_allocated()
+    out = flash_attn_xla(q, k, v, causal=True)
+    assert out.dtype == torch.float16
+    out.backward(torch.randn_like(out))
+    # the fp32 composite would materialize [b,h,s,s] scores (67 MB here
+    # per tensor, several live at once); the kernel path stays well under
+    peak_extra = torch.cuda.max_memory_allocated() - before
+    assert peak_extra < 40 * 2**20, f"{peak_extra/2**20:.1f} MiB"
+    from torchacc_amd.ops.flash_attn import _ref_attention
+    ref, _ = _ref_attention(q.detach().float().cpu(),
+                            k.detach().float().cpu(),
+                            v.detach().float().cpu(), d ** -0.5, True,
+                            (-1, -1))
+    err = (out.detach().float().cpu() - ref.float()).abs().max()
+    assert err < 3e-2, float(err)
+    assert q.grad is not None and q.grad.dtype == torch.float16
+
+
+def test_hf_lce_gpu_no_logits_and_parity():
+    """HF Llama on GPU: the patched causal-LM forward must not allocate
+    [b, s, vocab] logits and must match the unpatched loss."""
+    transformers = pytest.importorskip("transformers")
+    from transformers.models.llama.configuration_llama import LlamaConfig
+    from transformers.models.llama.modeling_llama import LlamaForCausalLM
+
+    import torchacc_amd  # noqa: F401 (import-time patch_fa)
+    from torchacc_amd.utils.patch import (_ORIG_CAUSAL_FWD,
+                                          apply_fused_kernel_patches)
+    apply_fused_kernel_patches()
+    cfg = LlamaConfig(
+        vocab_size=32000, hidden_size=512, intermediate_size=1024,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=4,
+        max_position_embeddings=2048,
+        attn_implementation="flash_attention_2")
+    torch.manual_seed(0)
+    with torch.device("cuda"):
+        model = LlamaForCausalLM(cfg).to(torch.bfloat16)
+    ids = torch.randint(0, 32000, (2, 2048), device="cuda")
+    torch.cuda.synchronize()
+    torch.cuda.reset_peak_memory_stats()
+    base = torch.cuda.memory_allocated()
+    out = model(input_ids=ids, labels=ids)
+    out.loss.backward()
+    peak = torch.cuda.max_memory_allocated() - base
+    assert out.logits is None
+    # full logits would be 2*2048*32000*2B = 250 MiB (+ fp32 copies in CE);
+    # the chunked path peaks far below that
+    assert peak < 180 * 2**20, f"{peak/2**20:.0f} MiB"
+    ref = _ORIG_CAUSAL_FWD[LlamaForCausalLM](model, input_ids=ids,
+                                             labels=ids)
+    assert abs(float(out.loss) - float(ref.loss)) < 5e-2
