@@ -730,17 +730,17 @@ std::vector<torch::Tensor> add_rmsnorm_backward(torch::Tensor dy,
   dim3 block(256);
   if (H == 4096) {
     hipLaunchKernelGGL(add_rmsnorm_bwd_fast<2>,
-                       dim3(std::min<long>(rows, 2048)), block, 0, stream,
+                       dim3(std::min<long>(rows, 512)), block, 0, stream,
                        dyp, drp, xp, wp, irp, dxp, dwp, (int)rows,
                        has_dresid);
   } else if (H == 8192) {
     hipLaunchKernelGGL(add_rmsnorm_bwd_fast<4>,
-                       dim3(std::min<long>(rows, 2048)), block, 0, stream,
+                       dim3(std::min<long>(rows, 512)), block, 0, stream,
                        dyp, drp, xp, wp, irp, dxp, dwp, (int)rows,
                        has_dresid);
   } else if (H == 2048) {
     hipLaunchKernelGGL(add_rmsnorm_bwd_fast<1>,
-                       dim3(std::min<long>(rows, 2048)), block, 0, stream,
+                       dim3(std::min<long>(rows, 512)), block, 0, stream,
                        dyp, drp, xp, wp, irp, dxp, dwp, (int)rows,
                        has_dresid);
   } else {
