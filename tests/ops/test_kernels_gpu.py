@@ -693,3 +693,33 @@ def test_hf_lce_gpu_no_logits_and_parity():
     assert peak < 0.7 * ref_peak, \
         f"chunked {peak/2**20:.0f} MiB vs full {ref_peak/2**20:.0f} MiB"
     assert abs(float(out.loss) - float(ref.loss)) < 5e-2
+
+
+@pytest.mark.parametrize("d", [96, 72, 48])
+def test_fa_padded_head_dims(d):
+    """Head dims outside {64,128} run zero-padded on the MFMA kernels
+    (reference pads and trims, ops/flash_attn.py:166-168)."""
+    from torchacc_amd.ops.flash_attn import flash_attn_xla, _ref_attention
+    torch.manual_seed(0)
+    b, s, h = 2, 256, 4
+    q = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    out = flash_attn_xla(q, k, v, causal=True)
+    assert out.shape[-1] == d
+    qr = q.detach().float().cpu().requires_grad_(True)
+    kr = k.detach().float().cpu().requires_grad_(True)
+    vr = v.detach().float().cpu().requires_grad_(True)
+    ref, _ = _ref_attention(qr, kr, vr, d ** -0.5, True, (-1, -1))
+    err = (out.detach().float().cpu() - ref).abs().max()
+    assert err < 2e-2, float(err)
+    g = torch.randn_like(out)
+    out.backward(g)
+    ref.backward(g.float().cpu())
+    for got, want in ((q.grad, qr.grad), (k.grad, kr.grad),
+                      (v.grad, vr.grad)):
+        e = (got.float().cpu() - want).abs().max()
+        assert e < 5e-2, float(e)

@@ -74,25 +74,41 @@ def _sac_push(entry):
 _warned_dims = set()
 
 
+def _pad_target(d: int):
+    """Kernel head_dim to zero-pad to (reference pads odd head dims and
+    trims in backward, ops/flash_attn.py:166-168); None -> composite."""
+    if d in (64, 128):
+        return None  # native
+    if d < 64:
+        return 64
+    if d < 128:
+        return 128
+    return -1  # unsupported: composite
+
+
 def _kernel_ext(q):
-    """The CDNA4 kernels cover bf16 with head_dim 64/128; anything else
-    runs the fp32 composite (warned once per dim, never silently on the
-    hot path — the flagship models are all head_dim 128)."""
+    """The CDNA4 kernels cover bf16 with head_dim 64/128 natively and any
+    d <= 128 via zero padding; d > 128 runs the fp32 composite (warned
+    once per dim — the flagship models are all head_dim 128)."""
     ext = dispatch(q)
     if ext is None:
         return None
     if q.dtype != torch.bfloat16:
         return None
     d = q.shape[-1]
-    if d not in (64, 128):
+    if _pad_target(d) == -1:
         if d not in _warned_dims:
             _warned_dims.add(d)
             from ..utils.logger import logger
             logger.warning(
-                "flash-attention head_dim %d unsupported by the CDNA4 "
-                "kernels (64/128); using the composite fallback", d)
+                "flash-attention head_dim %d > 128 unsupported by the "
+                "CDNA4 kernels; using the composite fallback", d)
         return None
     return ext
+
+
+def _pad_d(t: torch.Tensor, pad_to: int) -> torch.Tensor:
+    return torch.nn.functional.pad(t, (0, pad_to - t.shape[-1]))
 
 
 def _check_qkv(q, k, v):
@@ -177,11 +193,24 @@ class FlashAttnFunc(torch.autograd.Function):
                 # backward kernels regenerate the same keep-mask from it
                 seed = int(torch.randint(0, 2 ** 62, (1,)).item())
             if ext is not None:
-                out, lse = ext.fa_forward(
-                    q, k, v, softmax_scale, causal, wl, wr,
-                    q_lens if q_lens is not None else torch.empty(0),
-                    k_lens if k_lens is not None else torch.empty(0), al,
-                    dropout_p, seed)
+                d = q.shape[-1]
+                pad_to = _pad_target(d)
+                if pad_to:
+                    # zero-padded head dims: QK^T and the valid output dims
+                    # are unchanged; the pad columns of out are zero
+                    out, lse = ext.fa_forward(
+                        _pad_d(q, pad_to), _pad_d(k, pad_to),
+                        _pad_d(v, pad_to), softmax_scale, causal, wl, wr,
+                        q_lens if q_lens is not None else torch.empty(0),
+                        k_lens if k_lens is not None else torch.empty(0),
+                        al, dropout_p, seed)
+                    out = out[..., :d].contiguous()
+                else:
+                    out, lse = ext.fa_forward(
+                        q, k, v, softmax_scale, causal, wl, wr,
+                        q_lens if q_lens is not None else torch.empty(0),
+                        k_lens if k_lens is not None else torch.empty(0),
+                        al, dropout_p, seed)
             else:
                 out, lse = _ref_attention(q, k, v, softmax_scale, causal,
                                           (wl, wr), q_lens, k_lens,
@@ -208,12 +237,27 @@ class FlashAttnFunc(torch.autograd.Function):
         wl, wr = ctx.window
         if ext is not None:
             p_drop, seed = ctx.dropout
-            dq, dk, dv = ext.fa_backward(
-                dout, q, k, v, out, lse, ctx.softmax_scale, ctx.causal,
-                wl, wr,
-                q_lens if q_lens is not None else torch.empty(0),
-                k_lens if k_lens is not None else torch.empty(0), al,
-                p_drop, seed)
+            d = q.shape[-1]
+            pad_to = _pad_target(d)
+            if pad_to:
+                dq, dk, dv = ext.fa_backward(
+                    _pad_d(dout, pad_to), _pad_d(q, pad_to),
+                    _pad_d(k, pad_to), _pad_d(v, pad_to),
+                    _pad_d(out, pad_to), lse, ctx.softmax_scale,
+                    ctx.causal, wl, wr,
+                    q_lens if q_lens is not None else torch.empty(0),
+                    k_lens if k_lens is not None else torch.empty(0), al,
+                    p_drop, seed)
+                dq = dq[..., :d].contiguous()
+                dk = dk[..., :d].contiguous()
+                dv = dv[..., :d].contiguous()
+            else:
+                dq, dk, dv = ext.fa_backward(
+                    dout, q, k, v, out, lse, ctx.softmax_scale, ctx.causal,
+                    wl, wr,
+                    q_lens if q_lens is not None else torch.empty(0),
+                    k_lens if k_lens is not None else torch.empty(0), al,
+                    p_drop, seed)
         else:
             dq, dk, dv = _ref_fa_backward(dout, q, k, v, out, lse,
                                           ctx.softmax_scale, ctx.causal,
@@ -458,6 +502,14 @@ def _varlen_fwd_gpu(ext, q, k, v, cu_q, cu_k, softmax_scale, causal, wl,
     """Packed varlen: ONE fused kernel launch over the whole packed batch
     when q/k share the packing and there is no sliding window; otherwise
     one launch per sequence on the fixed-length kernel."""
+    d = q.shape[-1]
+    pad_to = _pad_target(d)
+    if pad_to:
+        out, lse = _varlen_fwd_gpu(ext, _pad_d(q, pad_to),
+                                   _pad_d(k, pad_to), _pad_d(v, pad_to),
+                                   cu_q, cu_k, softmax_scale, causal, wl,
+                                   wr)
+        return out[..., :d].contiguous(), lse
     total, h, _d = q.shape
     if _use_fused_varlen(q, k, cu_q, cu_k, wl, wr):
         bounds = _cu_to_bounds(cu_q.to(q.device), total)
@@ -483,6 +535,15 @@ def _varlen_fwd_gpu(ext, q, k, v, cu_q, cu_k, softmax_scale, causal, wl,
 
 def _varlen_bwd_gpu(ext, dout, q, k, v, out, lse, cu_q, cu_k, softmax_scale,
                     causal, wl, wr):
+    d = q.shape[-1]
+    pad_to = _pad_target(d)
+    if pad_to:
+        dq, dk, dv = _varlen_bwd_gpu(
+            ext, _pad_d(dout, pad_to), _pad_d(q, pad_to),
+            _pad_d(k, pad_to), _pad_d(v, pad_to), _pad_d(out, pad_to),
+            lse, cu_q, cu_k, softmax_scale, causal, wl, wr)
+        return (dq[..., :d].contiguous(), dk[..., :d].contiguous(),
+                dv[..., :d].contiguous())
     if _use_fused_varlen(q, k, cu_q, cu_k, wl, wr):
         bounds = _cu_to_bounds(cu_q.to(q.device), q.shape[0])
         return tuple(ext.fa_varlen_backward(dout, q, k, v, out, lse.contiguous(),
