@@ -723,3 +723,24 @@ def test_fa_padded_head_dims(d):
                       (v.grad, vr.grad)):
         e = (got.float().cpu() - want).abs().max()
         assert e < 5e-2, float(e)
+
+
+def test_cpu_offload_distinct_views_roundtrip():
+    """Two views sharing a storage base pointer but with different
+    geometry must offload/restore independently (advisor finding:
+    data_ptr-only dedupe returned the wrong tensor)."""
+    from torchacc_amd.utils.cpu_offload import \
+        AsyncDoubleBufferGroupOffloadHandler
+    h = AsyncDoubleBufferGroupOffloadHandler(num_offload_group=1)
+    base = torch.randn(64, 64, device="cuda")
+    v1 = base[:32]          # [32, 64], same data_ptr as base
+    v2 = base.reshape(-1)[:2048].view(16, 128)  # same ptr, new geometry
+    p1 = h.tensor_push(v1)
+    p2 = h.tensor_push(v2)
+    h.commit_group()
+    h.start_backward()
+    r2 = h.tensor_pop(p2)
+    r1 = h.tensor_pop(p1)
+    torch.cuda.synchronize()
+    assert r1.shape == v1.shape and torch.equal(r1, v1)
+    assert r2.shape == v2.shape and torch.equal(r2, v2)

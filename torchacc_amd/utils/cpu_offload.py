@@ -27,10 +27,10 @@ from .logger import logger
 class _OffloadGroup:
     def __init__(self, idx: int):
         self.idx = idx
-        self.packed: Dict[int, tuple] = {}   # data_ptr -> (host, meta)
+        self.packed: Dict[tuple, tuple] = {}  # view key -> (host, meta)
         self.d2h_event: Optional[torch.cuda.Event] = None
         self.h2d_event: Optional[torch.cuda.Event] = None
-        self.device_cache: Dict[int, torch.Tensor] = {}
+        self.device_cache: Dict[tuple, torch.Tensor] = {}
 
 
 class AsyncDoubleBufferGroupOffloadHandler:
@@ -55,7 +55,12 @@ class AsyncDoubleBufferGroupOffloadHandler:
                 not tensor.is_cuda or tensor.numel() < 1024:
             return ("keep", tensor)
         grp = self.groups.setdefault(gid, _OffloadGroup(gid))
-        key = tensor.data_ptr()
+        # dedupe key includes the VIEW's geometry: two distinct views of one
+        # storage (same data_ptr, different shape/stride) must offload
+        # separately or the pop returns the wrong tensor (the reference
+        # tracks per-tensor metadata, cpu_offload.py:310-519)
+        key = (tensor.data_ptr(), tuple(tensor.shape),
+               tuple(tensor.stride()), tensor.dtype)
         if key not in grp.packed:
             host = torch.empty(tensor.shape, dtype=tensor.dtype,
                                device="cpu", pin_memory=True)
