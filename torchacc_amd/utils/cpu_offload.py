@@ -7,8 +7,8 @@ TransformerEngine-derived) natively for HIP streams:
   HIP stream into pinned host buffers, grouped per layer-group commit;
 - during backward, groups are prefetched host->device ``num_prefetch_group``
   groups ahead on "offload_h2d";
-- tensors are deduped by data_ptr within a group (shared storages move
-  once);
+- tensors are deduped by (data_ptr, shape, stride, dtype) within a group
+  (identical views move once; distinct views of one storage move each);
 - user API: ``get_cpu_offload_context(num_offload_layers,
   num_prefetch_layers, ...)`` returning (context, sync_fn) — wrap each
   layer's forward in the context and call sync_fn between layer groups.
