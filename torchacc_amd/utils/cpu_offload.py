@@ -37,9 +37,16 @@ class AsyncDoubleBufferGroupOffloadHandler:
     """Bulk offload per layer group with double-buffered streams
     (reference :310-519)."""
 
-    def __init__(self, num_offload_group: int, num_prefetch_group: int = 1):
+    def __init__(self, num_offload_group: int, num_prefetch_group: int = 1,
+                 num_sync_group: int = 0):
         self.num_offload_group = num_offload_group
         self.num_prefetch_group = max(1, num_prefetch_group)
+        # first N groups offload synchronously (reference
+        # num_offload_sync_layers semantics, cpu_offload.py:521-605):
+        # their activations are guaranteed resident on host before the
+        # forward proceeds — bounding device memory exactly at the cost of
+        # overlap
+        self.num_sync_group = num_sync_group
         self.groups: Dict[int, _OffloadGroup] = {}
         self.current_group = 0
         self._on_gpu = torch.cuda.is_available()
@@ -54,6 +61,9 @@ class AsyncDoubleBufferGroupOffloadHandler:
         if not self._on_gpu or not self.offload_enabled(gid) or \
                 not tensor.is_cuda or tensor.numel() < 1024:
             return ("keep", tensor)
+        if gid < self.num_sync_group:
+            host = tensor.detach().cpu()
+            return ("sync_offloaded", host, tensor.device, tensor.dtype)
         grp = self.groups.setdefault(gid, _OffloadGroup(gid))
         # dedupe key includes the VIEW's geometry: two distinct views of one
         # storage (same data_ptr, different shape/stride) must offload
@@ -110,6 +120,8 @@ class AsyncDoubleBufferGroupOffloadHandler:
     def tensor_pop(self, packed) -> torch.Tensor:
         if packed[0] == "keep":
             return packed[1]
+        if packed[0] == "sync_offloaded":
+            return packed[1].to(packed[2])
         _, gid, key, shape, dtype = packed
         grp = self.groups[gid]
         # prefetch this group + the next ones toward group 0
@@ -180,12 +192,13 @@ def get_cpu_offload_context(num_offload_layers: int = 1,
                 x = layer(x)
             x = commit(x)    # marks the layer-group boundary
     """
-    if synchronous or num_offload_sync_layers > 0:
+    if synchronous:
         handler = SynchronizedGroupOffloadHandler(
             num_offload_layers, num_prefetch_layers)
     else:
         handler = AsyncDoubleBufferGroupOffloadHandler(
-            num_offload_layers, num_prefetch_layers)
+            num_offload_layers, num_prefetch_layers,
+            num_sync_group=num_offload_sync_layers)
 
     class _Commit(torch.autograd.Function):
         @staticmethod
