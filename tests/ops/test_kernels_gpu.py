@@ -744,3 +744,36 @@ def test_cpu_offload_distinct_views_roundtrip():
     torch.cuda.synchronize()
     assert r1.shape == v1.shape and torch.equal(r1, v1)
     assert r2.shape == v2.shape and torch.equal(r2, v2)
+
+
+@pytest.mark.parametrize("causal", [False, True])
+def test_fa_backward_bench_scale(causal):
+    """Regression guard at the BENCH shape scale (s=4096, d=128): a dkv
+    tiling change once passed every s<=1024 test while corrupting
+    gradients at s=4096 (bench loss diverged)."""
+    from torchacc_amd.ops._backend import require_extension
+    from torchacc_amd.ops.flash_attn import (_ref_attention,
+                                             _ref_fa_backward)
+    ext = require_extension()
+    b, s, h, hk, d = 1, 4096, 2, 2, 128
+    torch.manual_seed(0)
+    q = torch.randn(b, s, h, d)
+    k = torch.randn(b, s, hk, d)
+    v = torch.randn(b, s, hk, d)
+    dout = torch.randn(b, s, h, d)
+    scale = 1.0 / math.sqrt(d)
+    qg, kg, vg, dog = _to_gpu(q, k, v, dout)
+    o, lse = ext.fa_forward(qg, kg, vg, scale, causal, -1, -1,
+                            torch.empty(0), torch.empty(0),
+                            torch.empty(0), 0.0, 0)
+    dq, dk, dv = ext.fa_backward(dog, qg, kg, vg, o, lse, scale, causal,
+                                 -1, -1, torch.empty(0), torch.empty(0),
+                                 torch.empty(0), 0.0, 0)
+    ref_o, ref_lse = _ref_attention(q, k, v, scale, causal, (-1, -1))
+    rdq, rdk, rdv = _ref_fa_backward(dout, q, k, v, ref_o, ref_lse, scale,
+                                     causal, (-1, -1), None, None)
+    for name, got, want in (("dq", dq, rdq), ("dk", dk, rdk),
+                            ("dv", dv, rdv)):
+        err = (got.float().cpu() - want.float()).abs().max()
+        base = want.float().abs().max().clamp_min(1.0)
+        assert err / base < 4e-2, f"{name} err {float(err):.3f}"

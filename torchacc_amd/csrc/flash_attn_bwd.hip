@@ -72,9 +72,7 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
                        const int* __restrict__ k_lens) {
   constexpr int NT = D / 16;
   constexpr int NA = D / 32;
-  constexpr int QT = 64;            // q rows staged per barrier round
-  // (two 32-row sub-tiles computed sequentially per staging: same VGPR
-  // footprint as QT=32 but half the barriers and staging-loop overhead)
+  constexpr int QT = 32;            // q rows per staged tile
   constexpr int KVWG = 128;         // keys per workgroup (4 waves x 32)
   // transposed tiles [D][32]: 64B rows with a ((d>>3)&3)<<4 byte-XOR —
   // staging lanes differ in d by multiples of 8, so the XOR must use bits
@@ -202,14 +200,12 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
       }
       __syncthreads();
 
-      for (int sub = 0; sub < QT / 32; ++sub) {
-      const int q0s = q0 + sub * 32;
       // per-wave skip: this wave's keys [key_b, key_b+31] have no valid q
-      // in the sub-tile (causal diagonal / window-left bound) — staging
-      // and barriers are cooperative, so only the compute is guarded
+      // in the tile (causal diagonal / window-left bound) — staging and
+      // barriers are cooperative, so only the compute is guarded
       bool wave_active = true;
-      if (CAUSAL && q0s + 31 < key_b - shift) wave_active = false;
-      if (HAS_WINDOW && wl >= 0 && q0s > key_b + 31 - shift + wl)
+      if (CAUSAL && q0 + QT - 1 < key_b - shift) wave_active = false;
+      if (HAS_WINDOW && wl >= 0 && q0 > key_b + 31 - shift + wl)
         wave_active = false;
 
       if (wave_active) {
@@ -218,7 +214,7 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
       f32x16 dp = f32x16(0.f);
 #pragma unroll
       for (int t = 0; t < NT; ++t) {
-        const int row = sub * 32 + col;
+        const int row = col;
         unsigned byte = row * (D * 2) + (t * 16 + hi * 8) * 2;
         byte ^= (unsigned)((row & 7) << 4);
         bf16x8 qf = *reinterpret_cast<const bf16x8*>(
@@ -238,9 +234,9 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
       // (s reused as P, dp reused as dS — register budget)
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        const int qrow = q0s + CROW(r, hi);
-        const float lse_q = lse_lds[sub * 32 + CROW(r, hi)];
-        const float del_q = del_lds[sub * 32 + CROW(r, hi)];
+        const int qrow = q0 + CROW(r, hi);
+        const float lse_q = lse_lds[CROW(r, hi)];
+        const float del_q = del_lds[CROW(r, hi)];
         bool valid =
             (qrow < qlimit) && (mykey < klimit) && isfinite(lse_q);
         if (CAUSAL) valid &= (mykey <= qrow + shift);
@@ -271,7 +267,7 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
             "ds_read_b64_tr_b16 %3, %4 offset:%c8\n\t"                   \
             "s_waitcnt lgkmcnt(0)"                                         \
             : "=v"(ql_), "=v"(qh_), "=v"(dl_), "=v"(dh_)                   \
-            : "v"(tr_addr + (unsigned)(sub * 32 * D * 2)),                \
+            : "v"(tr_addr),                                                \
               "i"((tp_) * 16 * D * 2 + (a_) * 64),                         \
               "i"((tp_) * 16 * D * 2 + 4 * D * 2 + ((a_) ^ 1) * 64),       \
               "i"(QT * D * 2 + (tp_) * 16 * D * 2 + (a_) * 64),            \
@@ -312,7 +308,6 @@ void fa_bwd_dkv_kernel(const short* __restrict__ dOut,
       }
 #undef DKV_TR_STEP
       }  // wave_active
-      }  // sub
       __syncthreads();
     }
   }
@@ -571,7 +566,7 @@ static void launch_fa_bwd(const torch::Tensor& dout, const torch::Tensor& q,
       (short*)dq.data_ptr(), b, sq, sk, hq, hk, scale, wl, wr, qlp, klp
 
   dim3 gkv((sk + 127) / 128, b * hk), bkv(256);
-  const int lds_kv = (2 * 64 * D + 128 * D) * 2 + 2 * 64 * 4;
+  const int lds_kv = (2 * 32 * D + 128 * D) * 2 + 2 * 32 * 4;
   dim3 gq((sq + 255) / 256, b * hq), bq(512);
   const int lds_q = 3 * 128 * D * 2;
 
