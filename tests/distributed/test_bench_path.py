@@ -64,3 +64,23 @@ def test_bench_ulysses4_cpu_rehearsal():
 def test_bench_2d4_cpu_rehearsal():
     run_multiprocess(_bench_cp_worker, world_size=4, args=("2d",),
                      timeout=600)
+
+
+def _bench_fsdp_tp_worker(rank, world):
+    sys.path.insert(0, REPO_ROOT)
+    sys.argv = ["bench.py", "--model", "tiny", "--steps", "2",
+                "--warmup", "1", "--batch-size", "2", "--mode", "fsdp_tp",
+                "--tp", "2", "--gpus", str(world)]
+    import bench
+    buf = io.StringIO()
+    with contextlib.redirect_stdout(buf):
+        bench.main()
+    if rank == 0:
+        out = json.loads(buf.getvalue().strip().splitlines()[-1])
+        assert out["value"] > 0
+        assert out["config"]["parallelism"] == "fsdp2xtp2"
+
+
+def test_bench_fsdp_tp_cpu_rehearsal():
+    """The 70B FSDPxTP launch path (BASELINE config 4) at tp2 x fsdp2."""
+    run_multiprocess(_bench_fsdp_tp_worker, world_size=4, timeout=600)
