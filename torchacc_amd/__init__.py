@@ -110,3 +110,14 @@ __all__ = [
     "mark_dynamic", "get_global_context", "logger",
     "accelerate_hf_trainer", "llm",
 ]
+
+
+# import-time side effects (reference __init__.py:135-138): importing the
+# package already routes HF transformers' flash-attention entry point (and
+# its model-construction availability checks) to the CDNA4 kernels, so
+# `attn_implementation="flash_attention_2"` works without the CUDA
+# flash_attn package. TORCHACC_PATCH_FA=0 disables (reference patch.py:66).
+from .utils import patch as _patch  # noqa: E402
+
+_patch.patch_fa()
+_patch.patch_autocast()
