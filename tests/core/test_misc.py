@@ -209,3 +209,17 @@ def test_bucketing_pad_value_dict():
     assert batch["input_ids"].shape[-1] == 128
     assert (batch["input_ids"][:, 100:] == 0).all()
     assert (batch["labels"][:, 100:] == -100).all()
+
+
+def test_mem_plan_tool():
+    import subprocess
+    import sys
+    import os
+    root = os.path.dirname(os.path.dirname(
+        os.path.dirname(os.path.abspath(__file__))))
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, "tools", "mem_plan.py"),
+         "--model", "llama-2-70b", "--bs", "2", "--seq", "4096",
+         "--fsdp", "8"],
+        capture_output=True, text=True, check=True)
+    assert "recommended gc_cnt" in out.stdout
