@@ -115,7 +115,8 @@ class ColumnParallelLinear(nn.Module):
 
     def forward(self, x):
         x = copy_to_tp_region(x, self.group)
-        y = torch.nn.functional.linear(x, self.weight, self.bias)
+        from ..ops.linear import tuned_linear
+        y = tuned_linear(x, self.weight, self.bias)
         if self.gather_output and self.tp > 1:
             from ..ops.context_parallel.utils import \
                 gather_forward_split_backward
@@ -163,7 +164,8 @@ class RowParallelLinear(nn.Module):
             from ..ops.context_parallel.utils import \
                 split_forward_gather_backward
             x = split_forward_gather_backward(x, -1, self.group)
-        y = torch.nn.functional.linear(x, self.weight)
+        from ..ops.linear import tuned_linear
+        y = tuned_linear(x, self.weight)
         y = reduce_from_tp_region(y, self.group)
         if self.bias is not None:
             y = y + self.bias
