@@ -623,9 +623,9 @@ def test_plain_torch_scaler_is_syncfree():
     assert torch.isfinite(lin.weight.float()).all()
 
 
-def test_fp16_flash_attn_casts_to_bf16_kernels():
-    """fp16 q/k/v run the bf16 MFMA kernels (explicit cast), never the
-    fp32 composite (reference accepts fp16, ops/flash_attn.py:324-325)."""
+def test_fp16_flash_attn_native_kernels():
+    """fp16 q/k/v run the NATIVE f16 MFMA kernels (AttnElem traits), never
+    the fp32 composite (reference accepts fp16, ops/flash_attn.py:324-325)."""
     from torchacc_amd.ops.flash_attn import flash_attn_xla
     torch.manual_seed(0)
     b, s, h, d = 2, 256, 4, 128
@@ -777,3 +777,82 @@ def test_fa_backward_bench_scale(causal):
         err = (got.float().cpu() - want.float()).abs().max()
         base = want.float().abs().max().clamp_min(1.0)
         assert err / base < 4e-2, f"{name} err {float(err):.3f}"
+
+
+@pytest.mark.parametrize("causal", [False, True])
+def test_fa_fp16_forward_backward(causal):
+    """Native fp16 kernel family vs the fp32 composite: fwd + all grads."""
+    from torchacc_amd.ops._backend import require_extension
+    from torchacc_amd.ops.flash_attn import (_ref_attention,
+                                             _ref_fa_backward)
+    ext = require_extension()
+    b, s, h, hk, d = 2, 256, 8, 2, 128
+    torch.manual_seed(0)
+    q = torch.randn(b, s, h, d)
+    k = torch.randn(b, s, hk, d)
+    v = torch.randn(b, s, hk, d)
+    dout = torch.randn(b, s, h, d)
+    scale = 1.0 / math.sqrt(d)
+    qg, kg, vg, dog = _to_gpu(q, k, v, dout, dtype=torch.float16)
+    o, lse = ext.fa_forward(qg, kg, vg, scale, causal, -1, -1,
+                            torch.empty(0), torch.empty(0),
+                            torch.empty(0), 0.0, 0)
+    assert o.dtype == torch.float16
+    ref_o, ref_lse = _ref_attention(q, k, v, scale, causal, (-1, -1))
+    err = (o.float().cpu() - ref_o.float()).abs().max()
+    assert err < 2e-2, float(err)
+    dq, dk, dv = ext.fa_backward(dog, qg, kg, vg, o, lse, scale, causal,
+                                 -1, -1, torch.empty(0), torch.empty(0),
+                                 torch.empty(0), 0.0, 0)
+    rdq, rdk, rdv = _ref_fa_backward(dout, q, k, v, ref_o, ref_lse, scale,
+                                     causal, (-1, -1), None, None)
+    for name, got, want in (("dq", dq, rdq), ("dk", dk, rdk),
+                            ("dv", dv, rdv)):
+        e = (got.float().cpu() - want.float()).abs().max()
+        base = want.float().abs().max().clamp_min(1.0)
+        assert e / base < 4e-2, f"{name} err {float(e):.3f}"
+
+
+def test_fa_fp16_varlen_fused():
+    """fp16 through the fused packed-varlen kernels."""
+    from torchacc_amd.ops.flash_attn import (FlashAttnVarlenFunc,
+                                             _ref_varlen)
+    torch.manual_seed(0)
+    lens = [96, 160]
+    total, h, d = sum(lens), 4, 128
+    cu = torch.tensor([0, 96, 256], dtype=torch.int32, device="cuda")
+    q = torch.randn(total, h, d, device="cuda", dtype=torch.float16,
+                    requires_grad=True)
+    k = torch.randn(total, h, d, device="cuda", dtype=torch.float16,
+                    requires_grad=True)
+    v = torch.randn(total, h, d, device="cuda", dtype=torch.float16,
+                    requires_grad=True)
+    out, _ = FlashAttnVarlenFunc.apply(q, k, v, cu, cu, max(lens),
+                                       max(lens), 0.0, d ** -0.5, True,
+                                       (-1, -1), False)
+    assert out.dtype == torch.float16
+    ref, _ = _ref_varlen(q.detach().float().cpu(), k.detach().float().cpu(),
+                         v.detach().float().cpu(), cu.cpu(), cu.cpu(),
+                         d ** -0.5, True, (-1, -1))
+    err = (out.detach().float().cpu() - ref).abs().max()
+    assert err < 2e-2, float(err)
+    out.backward(torch.randn_like(out))
+    assert q.grad is not None and q.grad.dtype == torch.float16
+
+
+def test_fa_fp16_alibi():
+    """fp16 through the extra (alibi) kernel family."""
+    from torchacc_amd.ops.flash_attn import flash_attn_xla, _ref_attention
+    torch.manual_seed(0)
+    b, s, h, d = 2, 192, 4, 64
+    slopes = torch.tensor([0.5, 0.25, 0.125, 0.0625], device="cuda")
+    q = torch.randn(b, s, h, d, device="cuda", dtype=torch.float16)
+    k = torch.randn(b, s, h, d, device="cuda", dtype=torch.float16)
+    v = torch.randn(b, s, h, d, device="cuda", dtype=torch.float16)
+    out = flash_attn_xla(q, k, v, causal=True, alibi_slopes=slopes)
+    assert out.dtype == torch.float16
+    ref, _ = _ref_attention(q.float().cpu(), k.float().cpu(),
+                            v.float().cpu(), d ** -0.5, True, (-1, -1),
+                            alibi_slopes=slopes.cpu())
+    err = (out.float().cpu() - ref).abs().max()
+    assert err < 2e-2, float(err)

@@ -15,18 +15,64 @@ DEVINLINE unsigned attn_cvt_pk_bf16(float lo, float hi) {
   return r;
 }
 
+typedef _Float16 attn_f16x8 __attribute__((ext_vector_type(8)));
+typedef __fp16 attn_fp16x2 __attribute__((ext_vector_type(2)));
+
+// Element traits: the attention kernels are templated on FP16 so the SAME
+// MFMA/LDS structure serves bf16 (v_mfma_..._bf16, v_cvt_pk_bf16_f32) and
+// fp16 (v_mfma_..._f16, v_cvt_pkrtz_f16_f32). Storage stays short-based
+// (16-bit) everywhere — only the convert/mfma ops differ (reference
+// accepts fp16 and bf16, ops/flash_attn.py:324-325).
+template <bool FP16>
+struct AttnElem;
+
+template <>
+struct AttnElem<false> {  // bf16
+  static DEVINLINE float to_f32(short u) { return bf16_to_f32(u); }
+  static DEVINLINE short from_f32(float f) { return f32_to_bf16(f); }
+  static DEVINLINE unsigned cvt_pk(float lo, float hi) {
+    return attn_cvt_pk_bf16(lo, hi);
+  }
+  template <typename A, typename ACC>
+  static DEVINLINE ACC mfma(A a, A b, ACC acc) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
+  }
+};
+
+template <>
+struct AttnElem<true> {  // fp16
+  static DEVINLINE float to_f32(short u) {
+    __half h = __builtin_bit_cast(__half, u);
+    return __half2float(h);
+  }
+  static DEVINLINE short from_f32(float f) {
+    __half h = __float2half(f);
+    return __builtin_bit_cast(short, h);
+  }
+  static DEVINLINE unsigned cvt_pk(float lo, float hi) {
+    attn_fp16x2 h = __builtin_amdgcn_cvt_pkrtz(lo, hi);
+    return __builtin_bit_cast(unsigned, h);
+  }
+  template <typename A, typename ACC>
+  static DEVINLINE ACC mfma(A a, A b, ACC acc) {
+    attn_f16x8 af = __builtin_bit_cast(attn_f16x8, a);
+    attn_f16x8 bf = __builtin_bit_cast(attn_f16x8, b);
+    return __builtin_amdgcn_mfma_f32_32x32x16_f16(af, bf, acc, 0, 0, 0);
+  }
+};
+
 // T12 redistribution: given a 32-wide f32 accumulator pair layout
-// D[crow(r,hi)][col=lane&31], produce the bf16x8 B/A-fragment for mfma step
-// tp (16 rows [16*tp, 16*tp+16)): lane half hi receives rows 16tp+8hi+j.
-// Returns 4 packed u32 (8 bf16) in frag[0..3].
-template <typename V16>
+// D[crow(r,hi)][col=lane&31], produce the 16-bit x8 B/A-fragment for mfma
+// step tp (16 rows [16*tp, 16*tp+16)): lane half hi receives rows
+// 16tp+8hi+j. Returns 4 packed u32 (8 elems) in frag[0..3].
+template <typename ET, typename V16>
 DEVINLINE void t12_pack_frag(const V16& p16 /*16 regs*/, int tp,
                              unsigned* frag) {
 #pragma unroll
   for (int u = 0; u < 2; ++u) {
     const int r = 2 * u + 8 * tp;
-    unsigned va = attn_cvt_pk_bf16(p16[r], p16[r + 1]);
-    unsigned vb = attn_cvt_pk_bf16(p16[r + 4], p16[r + 5]);
+    unsigned va = ET::cvt_pk(p16[r], p16[r + 1]);
+    unsigned vb = ET::cvt_pk(p16[r + 4], p16[r + 5]);
     auto sw = __builtin_amdgcn_permlane32_swap(va, vb, false, false);
     frag[u] = sw[0];
     frag[u + 2] = sw[1];

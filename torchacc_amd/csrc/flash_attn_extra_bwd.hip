@@ -34,6 +34,7 @@
 // ---------------------------------------------------------------------------
 // 1. preprocess: delta = rowsum(dO * O)   [b,h,sq] fp32
 // ---------------------------------------------------------------------------
+template <bool FP16>
 __global__ void fa_bwd_preprocess_kernel_x(const short* __restrict__ dO,
                                          const short* __restrict__ O,
                                          float* __restrict__ delta, int b,
@@ -47,8 +48,9 @@ __global__ void fa_bwd_preprocess_kernel_x(const short* __restrict__ dO,
   const short* orow = O + row * D;
   float acc = 0.f;
   for (int i = lane * 2; i < D; i += WAVE * 2) {
-    acc += bf16_to_f32(dr[i]) * bf16_to_f32(orow[i]) +
-           bf16_to_f32(dr[i + 1]) * bf16_to_f32(orow[i + 1]);
+    acc += AttnElem<FP16>::to_f32(dr[i]) * AttnElem<FP16>::to_f32(orow[i]) +
+           AttnElem<FP16>::to_f32(dr[i + 1]) *
+               AttnElem<FP16>::to_f32(orow[i + 1]);
   }
   acc = wave_reduce_sum(acc);
   if (lane == 0) {
@@ -62,7 +64,8 @@ __global__ void fa_bwd_preprocess_kernel_x(const short* __restrict__ dO,
 // ---------------------------------------------------------------------------
 // 2. dK/dV kernel
 // ---------------------------------------------------------------------------
-template <int D, bool CAUSAL, bool HAS_WINDOW, bool HAS_LENS, bool HAS_EXTRA>
+template <int D, bool CAUSAL, bool HAS_WINDOW, bool HAS_LENS, bool HAS_EXTRA,
+          bool FP16>
 __global__ __launch_bounds__(256, 2)
 void fa_bwd_dkv_kernel_x(const short* __restrict__ dOut,
                        const short* __restrict__ Q,
@@ -227,9 +230,8 @@ void fa_bwd_dkv_kernel_x(const short* __restrict__ dOut,
         vbyte ^= (unsigned)((vrow & 7) << 4);
         bf16x8 vf = *reinterpret_cast<const bf16x8*>(
             reinterpret_cast<const char*>(v_lds) + vbyte);
-        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfrag[t], s, 0, 0,
-                                                    0);
-        dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(df, vf, dp, 0, 0, 0);
+        s = AttnElem<FP16>::mfma(qf, kfrag[t], s);
+        dp = AttnElem<FP16>::mfma(df, vf, dp);
       }
       // P = exp(S*scale - lse); dS = P*(dP - delta)*scale
       // (s reused as P, dp reused as dS — register budget)
@@ -284,16 +286,14 @@ void fa_bwd_dkv_kernel_x(const short* __restrict__ dOut,
         attn_u32x4 ud_ = {dl_.x, dl_.y, dh_.x, dh_.y};                     \
         bf16x8 qbf = __builtin_bit_cast(bf16x8, uq_);                      \
         bf16x8 dob = __builtin_bit_cast(bf16x8, ud_);                      \
-        dvacc[a_] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(               \
-            pb, dob, dvacc[a_], 0, 0, 0);                                  \
-        dkacc[a_] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(               \
-            dsb, qbf, dkacc[a_], 0, 0, 0);                                 \
+        dvacc[a_] = AttnElem<FP16>::mfma(pb, dob, dvacc[a_]);              \
+        dkacc[a_] = AttnElem<FP16>::mfma(dsb, qbf, dkacc[a_]);             \
       }
 #pragma unroll
       for (int tp = 0; tp < 2; ++tp) {
         unsigned pfr[4], dsfr[4];
-        t12_pack_frag(s, tp, pfr);
-        t12_pack_frag(dp, tp, dsfr);
+        t12_pack_frag<AttnElem<FP16>>(s, tp, pfr);
+        t12_pack_frag<AttnElem<FP16>>(dp, tp, dsfr);
         bf16x8 pb = *reinterpret_cast<const bf16x8*>(pfr);
         bf16x8 dsb = *reinterpret_cast<const bf16x8*>(dsfr);
         if (tp == 0) {
@@ -326,8 +326,8 @@ void fa_bwd_dkv_kernel_x(const short* __restrict__ dOut,
       const int d = a * 32 + col;
       if (key < sk) {
         const long base = ((long)b * sk + key) * hk * D + (long)kh * D;
-        dK[base + d] = f32_to_bf16(dkacc[a][r]);
-        dV[base + d] = f32_to_bf16(dvacc[a][r]);
+        dK[base + d] = AttnElem<FP16>::from_f32(dkacc[a][r]);
+        dV[base + d] = AttnElem<FP16>::from_f32(dvacc[a][r]);
       }
     }
   }
@@ -336,7 +336,8 @@ void fa_bwd_dkv_kernel_x(const short* __restrict__ dOut,
 // ---------------------------------------------------------------------------
 // 3. dQ kernel (forward structure)
 // ---------------------------------------------------------------------------
-template <int D, bool CAUSAL, bool HAS_WINDOW, bool HAS_LENS, bool HAS_EXTRA>
+template <int D, bool CAUSAL, bool HAS_WINDOW, bool HAS_LENS, bool HAS_EXTRA,
+          bool FP16>
 __global__ __launch_bounds__(512, 2)
 void fa_bwd_dq_kernel_x(const short* __restrict__ dOut,
                       const short* __restrict__ Q,
@@ -480,10 +481,8 @@ void fa_bwd_dq_kernel_x(const short* __restrict__ dOut,
               reinterpret_cast<const char*>(k_lds) + byte);
           bf16x8 vf = *reinterpret_cast<const bf16x8*>(
               reinterpret_cast<const char*>(v_lds) + byte);
-          s[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfrag[t],
-                                                          s[kb], 0, 0, 0);
-          dp[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dofrag[t],
-                                                           dp[kb], 0, 0, 0);
+          s[kb] = AttnElem<FP16>::mfma(kf, qfrag[t], s[kb]);
+          dp[kb] = AttnElem<FP16>::mfma(vf, dofrag[t], dp[kb]);
         }
       }
       // reuse dp as dS (register budget)
@@ -518,7 +517,7 @@ void fa_bwd_dq_kernel_x(const short* __restrict__ dOut,
 #pragma unroll
         for (int tp = 0; tp < 2; ++tp) {
           unsigned dsfr[4];
-          t12_pack_frag(dp[kb], tp, dsfr);
+          t12_pack_frag<AttnElem<FP16>>(dp[kb], tp, dsfr);
           bf16x8 dsb = *reinterpret_cast<const bf16x8*>(dsfr);
 #pragma unroll
           for (int a = 0; a < NA; ++a) {
@@ -528,8 +527,7 @@ void fa_bwd_dq_kernel_x(const short* __restrict__ dOut,
             byte ^= (unsigned)(((d >> 3) & 7) << 4);
             bf16x8 ktb = *reinterpret_cast<const bf16x8*>(
                 reinterpret_cast<const char*>(kt_lds) + byte);
-            dqacc[a] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                dsb, ktb, dqacc[a], 0, 0, 0);
+            dqacc[a] = AttnElem<FP16>::mfma(dsb, ktb, dqacc[a]);
           }
         }
       }
@@ -546,7 +544,8 @@ void fa_bwd_dq_kernel_x(const short* __restrict__ dOut,
       const int d = a * 32 + col;
       if (q_r < sq) {
         const long obase = ((long)(b * sq + q_r) * hq + h) * D;
-        dQ[obase + d] = (q_r < qlimit) ? f32_to_bf16(dqacc[a][r]) : (short)0;
+        dQ[obase + d] = (q_r < qlimit)
+            ? AttnElem<FP16>::from_f32(dqacc[a][r]) : (short)0;
       }
     }
   }
@@ -556,7 +555,7 @@ void fa_bwd_dq_kernel_x(const short* __restrict__ dOut,
 // host wrappers
 // ---------------------------------------------------------------------------
 
-template <int D>
+template <int D, bool FP16>
 static void launch_fa_bwd_x(const torch::Tensor& dout, const torch::Tensor& q,
                           const torch::Tensor& k, const torch::Tensor& v,
                           const torch::Tensor& lse,
@@ -591,10 +590,10 @@ static void launch_fa_bwd_x(const torch::Tensor& dout, const torch::Tensor& q,
 
 #define DISPATCH_CASE(C, W, L, DR)                                          \
   if (causal == C && has_window == W && has_lens == L && has_drop == DR) {   \
-    hipLaunchKernelGGL((fa_bwd_dkv_kernel_x<D, C, W, L, DR>), gkv, bkv,        \
-                       lds_kv, stream, ARGS_DKV);                            \
-    hipLaunchKernelGGL((fa_bwd_dq_kernel_x<D, C, W, L, DR>), gq, bq, lds_q,    \
-                       stream, ARGS_DQ);                                     \
+    hipLaunchKernelGGL((fa_bwd_dkv_kernel_x<D, C, W, L, DR, FP16>), gkv,    \
+                       bkv, lds_kv, stream, ARGS_DKV);                       \
+    hipLaunchKernelGGL((fa_bwd_dq_kernel_x<D, C, W, L, DR, FP16>), gq, bq,   \
+                       lds_q, stream, ARGS_DQ);                              \
     return;                                                                  \
   }
   const bool has_drop = true;  // extra TU: alibi/dropout active
@@ -619,7 +618,8 @@ std::vector<torch::Tensor> fa_backward_extra(torch::Tensor dout, torch::Tensor q
                                        torch::Tensor k_lens,
                                        torch::Tensor alibi_slopes,
                                        double p_drop, long rng_seed) {
-  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16,
+  const bool fp16 = q.scalar_type() == torch::kHalf;
+  TORCH_CHECK(q.is_cuda() && (q.scalar_type() == torch::kBFloat16 || fp16),
               "fa_backward: bf16 only");
   TORCH_CHECK(dout.is_contiguous() && q.is_contiguous() &&
               k.is_contiguous() && v.is_contiguous());
@@ -631,10 +631,18 @@ std::vector<torch::Tensor> fa_backward_extra(torch::Tensor dout, torch::Tensor q
   {
     const long rows = (long)b * sq * hq;
     const long grid = (rows + 3) / 4;
-    hipLaunchKernelGGL(fa_bwd_preprocess_kernel_x, dim3((unsigned)grid),
-                       dim3(256), 0, stream, (const short*)dout.data_ptr(),
-                       (const short*)out.data_ptr(), delta.data_ptr<float>(),
-                       b, sq, hq, D);
+    if (fp16)
+      hipLaunchKernelGGL(fa_bwd_preprocess_kernel_x<true>,
+                         dim3((unsigned)grid), dim3(256), 0, stream,
+                         (const short*)dout.data_ptr(),
+                         (const short*)out.data_ptr(),
+                         delta.data_ptr<float>(), b, sq, hq, D);
+    else
+      hipLaunchKernelGGL(fa_bwd_preprocess_kernel_x<false>,
+                         dim3((unsigned)grid), dim3(256), 0, stream,
+                         (const short*)dout.data_ptr(),
+                         (const short*)out.data_ptr(),
+                         delta.data_ptr<float>(), b, sq, hq, D);
   }
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
@@ -648,15 +656,27 @@ std::vector<torch::Tensor> fa_backward_extra(torch::Tensor dout, torch::Tensor q
                 : alibi_slopes;
   const float* alp = al.numel() ? al.data_ptr<float>() : nullptr;
   if (D == 128) {
-    launch_fa_bwd_x<128>(dout, q, k, v, lse_c, delta, dq, dk, dv,
-                       (float)softmax_scale, causal, (int)wl, (int)wr, qlp,
-                       klp, alp, (float)p_drop,
-                       (unsigned long long)rng_seed, stream);
+    if (fp16)
+      launch_fa_bwd_x<128, true>(dout, q, k, v, lse_c, delta, dq, dk, dv,
+                                 (float)softmax_scale, causal, (int)wl,
+                                 (int)wr, qlp, klp, alp, (float)p_drop,
+                                 (unsigned long long)rng_seed, stream);
+    else
+      launch_fa_bwd_x<128, false>(dout, q, k, v, lse_c, delta, dq, dk, dv,
+                                  (float)softmax_scale, causal, (int)wl,
+                                  (int)wr, qlp, klp, alp, (float)p_drop,
+                                  (unsigned long long)rng_seed, stream);
   } else {
-    launch_fa_bwd_x<64>(dout, q, k, v, lse_c, delta, dq, dk, dv,
-                      (float)softmax_scale, causal, (int)wl, (int)wr, qlp,
-                      klp, alp, (float)p_drop,
-                      (unsigned long long)rng_seed, stream);
+    if (fp16)
+      launch_fa_bwd_x<64, true>(dout, q, k, v, lse_c, delta, dq, dk, dv,
+                                (float)softmax_scale, causal, (int)wl,
+                                (int)wr, qlp, klp, alp, (float)p_drop,
+                                (unsigned long long)rng_seed, stream);
+    else
+      launch_fa_bwd_x<64, false>(dout, q, k, v, lse_c, delta, dq, dk, dv,
+                                 (float)softmax_scale, causal, (int)wl,
+                                 (int)wr, qlp, klp, alp, (float)p_drop,
+                                 (unsigned long long)rng_seed, stream);
   }
   HIP_CHECK_LAST();
   return {dq, dk, dv};

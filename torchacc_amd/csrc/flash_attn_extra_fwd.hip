@@ -26,11 +26,12 @@
 #include "attn_common.h"
 
 typedef float f32x16_ __attribute__((ext_vector_type(16)));
-#define cvt_pk_bf16 attn_cvt_pk_bf16
+#define cvt_pk_bf16 attn_cvt_pk_bf16  // bf16 path alias
 
 // HAS_EXTRA: alibi and/or dropout active (kept out of the
 // register budget of the vanilla path)
-template <int D, bool CAUSAL, bool HAS_WINDOW, bool HAS_LENS, bool HAS_EXTRA>
+template <int D, bool CAUSAL, bool HAS_WINDOW, bool HAS_LENS, bool HAS_EXTRA,
+          bool FP16>
 __global__ __launch_bounds__(512, 2)
 void fa_fwd_kernel_x(const short* __restrict__ Q, const short* __restrict__ K,
                    const short* __restrict__ V, short* __restrict__ O,
@@ -203,8 +204,7 @@ void fa_fwd_kernel_x(const short* __restrict__ Q, const short* __restrict__ K,
           byte ^= (unsigned)((row & 7) << 4);
           bf16x8 kf = *reinterpret_cast<const bf16x8*>(
               reinterpret_cast<const char*>(kbuf) + byte);
-          p[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfrag[t],
-                                                          p[kb], 0, 0, 0);
+          p[kb] = AttnElem<FP16>::mfma(kf, qfrag[t], p[kb]);
         }
       }
       // ---- mask + scale --------------------------------------------------
@@ -287,8 +287,9 @@ void fa_fwd_kernel_x(const short* __restrict__ Q, const short* __restrict__ K,
 #pragma unroll
           for (int u = 0; u < 2; ++u) {
             const int r = 2 * u + 8 * tp;
-            unsigned va = cvt_pk_bf16(p[kb][r], p[kb][r + 1]);
-            unsigned vb = cvt_pk_bf16(p[kb][r + 4], p[kb][r + 5]);
+            unsigned va = AttnElem<FP16>::cvt_pk(p[kb][r], p[kb][r + 1]);
+            unsigned vb = AttnElem<FP16>::cvt_pk(p[kb][r + 4],
+                                                 p[kb][r + 5]);
             auto sw = __builtin_amdgcn_permlane32_swap(va, vb, false, false);
             pb[kb * 2 + tp][u] = sw[0];
             pb[kb * 2 + tp][u + 2] = sw[1];
@@ -307,8 +308,7 @@ void fa_fwd_kernel_x(const short* __restrict__ Q, const short* __restrict__ K,
           bf16x8 vf = *reinterpret_cast<const bf16x8*>(
               reinterpret_cast<const char*>(vbuf) + byte);
           bf16x8 pf = *reinterpret_cast<const bf16x8*>(&pb[st][0]);
-          oacc[a] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pf, oacc[a],
-                                                            0, 0, 0);
+          oacc[a] = AttnElem<FP16>::mfma(vf, pf, oacc[a]);
         }
       }
     }
@@ -332,7 +332,7 @@ void fa_fwd_kernel_x(const short* __restrict__ Q, const short* __restrict__ K,
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int d = a * 32 + CROW(r, hi);
-        O[obase + d] = f32_to_bf16(oacc[a][r] * inv_l);
+        O[obase + d] = AttnElem<FP16>::from_f32(oacc[a][r] * inv_l);
       }
     }
     if (hi == 0) {
@@ -358,7 +358,7 @@ void fa_fwd_kernel_x(const short* __restrict__ Q, const short* __restrict__ K,
 // host wrapper
 // ---------------------------------------------------------------------------
 
-template <int D>
+template <int D, bool FP16>
 static void launch_fa_fwd_x(const torch::Tensor& q, const torch::Tensor& k,
                           const torch::Tensor& v, torch::Tensor& o,
                           torch::Tensor& lse, float scale, bool causal,
@@ -379,8 +379,8 @@ static void launch_fa_fwd_x(const torch::Tensor& q, const torch::Tensor& k,
 
 #define LAUNCH_CASE(C, W, L, DR)                                            \
   if (causal == C && has_window == W && has_lens == L && has_drop == DR) {   \
-    hipLaunchKernelGGL((fa_fwd_kernel_x<D, C, W, L, DR>), grid, block, lds,    \
-                       stream, (const short*)q.data_ptr(),                   \
+    hipLaunchKernelGGL((fa_fwd_kernel_x<D, C, W, L, DR, FP16>), grid,       \
+                       block, lds, stream, (const short*)q.data_ptr(),       \
                        (const short*)k.data_ptr(),                           \
                        (const short*)v.data_ptr(), (short*)o.data_ptr(),     \
                        lse.data_ptr<float>(), b, sq, sk, hq, hk, scale, wl,  \
@@ -408,8 +408,10 @@ std::vector<torch::Tensor> fa_forward_extra(torch::Tensor q, torch::Tensor k,
                                       double p_drop, long rng_seed) {
   TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
               v.is_contiguous());
-  TORCH_CHECK(q.scalar_type() == torch::kBFloat16,
-              "fa_forward: bf16 only (CDNA4 MFMA path)");
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16 ||
+                  q.scalar_type() == torch::kHalf,
+              "fa_forward: bf16/fp16 only (CDNA4 MFMA path)");
+  const bool fp16 = q.scalar_type() == torch::kHalf;
   const int D = q.size(3);
   TORCH_CHECK(D == 64 || D == 128, "fa_forward: head_dim must be 64 or 128");
   TORCH_CHECK(q.size(2) % k.size(2) == 0);
@@ -427,13 +429,27 @@ std::vector<torch::Tensor> fa_forward_extra(torch::Tensor q, torch::Tensor k,
               "alibi_slopes must be [num_heads]");
   auto stream = at::hip::getCurrentHIPStream();
   if (D == 128) {
-    launch_fa_fwd_x<128>(q, k, v, o, lse, (float)softmax_scale, causal,
-                       (int)wl, (int)wr, ql, kl, al, (float)p_drop,
-                       (unsigned long long)rng_seed, stream);
+    if (fp16)
+      launch_fa_fwd_x<128, true>(q, k, v, o, lse, (float)softmax_scale,
+                                 causal, (int)wl, (int)wr, ql, kl, al,
+                                 (float)p_drop, (unsigned long long)rng_seed,
+                                 stream);
+    else
+      launch_fa_fwd_x<128, false>(q, k, v, o, lse, (float)softmax_scale,
+                                  causal, (int)wl, (int)wr, ql, kl, al,
+                                  (float)p_drop,
+                                  (unsigned long long)rng_seed, stream);
   } else {
-    launch_fa_fwd_x<64>(q, k, v, o, lse, (float)softmax_scale, causal, (int)wl,
-                      (int)wr, ql, kl, al, (float)p_drop,
-                      (unsigned long long)rng_seed, stream);
+    if (fp16)
+      launch_fa_fwd_x<64, true>(q, k, v, o, lse, (float)softmax_scale,
+                                causal, (int)wl, (int)wr, ql, kl, al,
+                                (float)p_drop, (unsigned long long)rng_seed,
+                                stream);
+    else
+      launch_fa_fwd_x<64, false>(q, k, v, o, lse, (float)softmax_scale,
+                                 causal, (int)wl, (int)wr, ql, kl, al,
+                                 (float)p_drop, (unsigned long long)rng_seed,
+                                 stream);
   }
   HIP_CHECK_LAST();
   return {o, lse};

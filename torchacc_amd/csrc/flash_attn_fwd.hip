@@ -16,26 +16,18 @@
 //     epilogue normalizes by 1/l and stores out + logsumexp.
 //
 // Covers: causal (bottom-right aligned), GQA (h_kv | h), sliding window,
-// per-batch varlen (q_lens/k_lens), D in {64, 128}. bf16 only.
+// per-batch varlen (q_lens/k_lens), D in {64, 128}; bf16 and fp16
+// (AttnElem traits: same MFMA ladder, v_mfma_..._f16 / v_cvt_pkrtz).
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include "common.h"
+#include "attn_common.h"
 
 #define LOG2E 1.4426950408889634f
 
 typedef float f32x16_ __attribute__((ext_vector_type(16)));
 
-DEVINLINE unsigned cvt_pk_bf16(float lo, float hi) {
-  unsigned r;
-  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
-               : "=v"(r) : "v"(lo), "v"(hi));
-  return r;
-}
-
-// crow: D-matrix row for accumulator register r on lane-half hi (32x32 mfma)
-#define CROW(r, hi) (((r) & 3) + 8 * ((r) >> 2) + 4 * (hi))
-
-template <int D, bool CAUSAL, bool HAS_WINDOW, bool HAS_LENS>
+template <int D, bool CAUSAL, bool HAS_WINDOW, bool HAS_LENS, bool FP16>
 __global__ __launch_bounds__(512, 2)
 void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
                    const short* __restrict__ V, short* __restrict__ O,
@@ -199,8 +191,7 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
           byte ^= (unsigned)((row & 7) << 4);
           bf16x8 kf = *reinterpret_cast<const bf16x8*>(
               reinterpret_cast<const char*>(kbuf) + byte);
-          p[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfrag[t],
-                                                          p[kb], 0, 0, 0);
+          p[kb] = AttnElem<FP16>::mfma(kf, qfrag[t], p[kb]);
         }
       }
       // ---- mask + scale --------------------------------------------------
@@ -266,8 +257,9 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
 #pragma unroll
           for (int u = 0; u < 2; ++u) {
             const int r = 2 * u + 8 * tp;
-            unsigned va = cvt_pk_bf16(p[kb][r], p[kb][r + 1]);
-            unsigned vb = cvt_pk_bf16(p[kb][r + 4], p[kb][r + 5]);
+            unsigned va = AttnElem<FP16>::cvt_pk(p[kb][r], p[kb][r + 1]);
+            unsigned vb = AttnElem<FP16>::cvt_pk(p[kb][r + 4],
+                                                 p[kb][r + 5]);
             auto sw = __builtin_amdgcn_permlane32_swap(va, vb, false, false);
             pb[kb * 2 + tp][u] = sw[0];
             pb[kb * 2 + tp][u + 2] = sw[1];
@@ -286,8 +278,7 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
           bf16x8 vf = *reinterpret_cast<const bf16x8*>(
               reinterpret_cast<const char*>(vbuf) + byte);
           bf16x8 pf = *reinterpret_cast<const bf16x8*>(&pb[st][0]);
-          oacc[a] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pf, oacc[a],
-                                                            0, 0, 0);
+          oacc[a] = AttnElem<FP16>::mfma(vf, pf, oacc[a]);
         }
       }
     }
@@ -311,7 +302,7 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int d = a * 32 + CROW(r, hi);
-        O[obase + d] = f32_to_bf16(oacc[a][r] * inv_l);
+        O[obase + d] = AttnElem<FP16>::from_f32(oacc[a][r] * inv_l);
       }
     }
     if (hi == 0) {
@@ -337,7 +328,7 @@ void fa_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
 // host wrapper
 // ---------------------------------------------------------------------------
 
-template <int D>
+template <int D, bool FP16>
 static void launch_fa_fwd(const torch::Tensor& q, const torch::Tensor& k,
                           const torch::Tensor& v, torch::Tensor& o,
                           torch::Tensor& lse, float scale, bool causal,
@@ -354,8 +345,8 @@ static void launch_fa_fwd(const torch::Tensor& q, const torch::Tensor& k,
   const int* klp = k_lens.numel() ? k_lens.data_ptr<int>() : nullptr;
 
 #define LAUNCH(CAUSAL, WIN, LENS)                                            \
-  hipLaunchKernelGGL((fa_fwd_kernel<D, CAUSAL, WIN, LENS>), grid, block,     \
-                     lds, stream, (const short*)q.data_ptr(),                \
+  hipLaunchKernelGGL((fa_fwd_kernel<D, CAUSAL, WIN, LENS, FP16>), grid,      \
+                     block, lds, stream, (const short*)q.data_ptr(),         \
                      (const short*)k.data_ptr(),                             \
                      (const short*)v.data_ptr(), (short*)o.data_ptr(),       \
                      lse.data_ptr<float>(), b, sq, sk, hq, hk, scale, wl,    \
@@ -390,8 +381,10 @@ std::vector<torch::Tensor> fa_forward(torch::Tensor q, torch::Tensor k,
   }
   TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
               v.is_contiguous());
-  TORCH_CHECK(q.scalar_type() == torch::kBFloat16,
-              "fa_forward: bf16 only (CDNA4 MFMA path)");
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16 ||
+                  q.scalar_type() == torch::kHalf,
+              "fa_forward: bf16/fp16 only (CDNA4 MFMA path)");
+  const bool fp16 = q.scalar_type() == torch::kHalf;
   const int D = q.size(3);
   TORCH_CHECK(D == 64 || D == 128, "fa_forward: head_dim must be 64 or 128");
   TORCH_CHECK(q.size(2) % k.size(2) == 0);
@@ -404,11 +397,19 @@ std::vector<torch::Tensor> fa_forward(torch::Tensor q, torch::Tensor k,
                            : k_lens;
   auto stream = at::hip::getCurrentHIPStream();
   if (D == 128) {
-    launch_fa_fwd<128>(q, k, v, o, lse, (float)softmax_scale, causal,
-                       (int)wl, (int)wr, ql, kl, stream);
+    if (fp16)
+      launch_fa_fwd<128, true>(q, k, v, o, lse, (float)softmax_scale,
+                               causal, (int)wl, (int)wr, ql, kl, stream);
+    else
+      launch_fa_fwd<128, false>(q, k, v, o, lse, (float)softmax_scale,
+                                causal, (int)wl, (int)wr, ql, kl, stream);
   } else {
-    launch_fa_fwd<64>(q, k, v, o, lse, (float)softmax_scale, causal, (int)wl,
-                      (int)wr, ql, kl, stream);
+    if (fp16)
+      launch_fa_fwd<64, true>(q, k, v, o, lse, (float)softmax_scale, causal,
+                              (int)wl, (int)wr, ql, kl, stream);
+    else
+      launch_fa_fwd<64, false>(q, k, v, o, lse, (float)softmax_scale,
+                               causal, (int)wl, (int)wr, ql, kl, stream);
   }
   HIP_CHECK_LAST();
   return {o, lse};

@@ -25,7 +25,7 @@ typedef float f32x16_ __attribute__((ext_vector_type(16)));
 // ---------------------------------------------------------------------------
 // 1. forward
 // ---------------------------------------------------------------------------
-template <int D, bool CAUSAL>
+template <int D, bool CAUSAL, bool FP16>
 __global__ __launch_bounds__(512, 2)
 void fa_vl_fwd_kernel(const short* __restrict__ Q,
                       const short* __restrict__ K,
@@ -167,8 +167,7 @@ void fa_vl_fwd_kernel(const short* __restrict__ Q,
           byte ^= (unsigned)((row & 7) << 4);
           bf16x8 kf = *reinterpret_cast<const bf16x8*>(
               reinterpret_cast<const char*>(kbuf) + byte);
-          p[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfrag[t],
-                                                          p[kb], 0, 0, 0);
+          p[kb] = AttnElem<FP16>::mfma(kf, qfrag[t], p[kb]);
         }
       }
       // full = every (row, key) of the wave inside one sequence span:
@@ -224,8 +223,9 @@ void fa_vl_fwd_kernel(const short* __restrict__ Q,
 #pragma unroll
           for (int u = 0; u < 2; ++u) {
             const int r = 2 * u + 8 * tp;
-            unsigned va = attn_cvt_pk_bf16(p[kb][r], p[kb][r + 1]);
-            unsigned vb = attn_cvt_pk_bf16(p[kb][r + 4], p[kb][r + 5]);
+            unsigned va = AttnElem<FP16>::cvt_pk(p[kb][r], p[kb][r + 1]);
+            unsigned vb = AttnElem<FP16>::cvt_pk(p[kb][r + 4],
+                                                 p[kb][r + 5]);
             auto sw = __builtin_amdgcn_permlane32_swap(va, vb, false, false);
             pb[kb * 2 + tp][u] = sw[0];
             pb[kb * 2 + tp][u + 2] = sw[1];
@@ -242,8 +242,7 @@ void fa_vl_fwd_kernel(const short* __restrict__ Q,
           bf16x8 vf = *reinterpret_cast<const bf16x8*>(
               reinterpret_cast<const char*>(vbuf) + byte);
           bf16x8 pf = *reinterpret_cast<const bf16x8*>(&pb[st][0]);
-          oacc[a] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pf, oacc[a],
-                                                            0, 0, 0);
+          oacc[a] = AttnElem<FP16>::mfma(vf, pf, oacc[a]);
         }
       }
     }
@@ -263,7 +262,7 @@ void fa_vl_fwd_kernel(const short* __restrict__ Q,
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int d = a * 32 + CROW(r, hi);
-        O[obase + d] = f32_to_bf16(oacc[a][r] * inv_l);
+        O[obase + d] = AttnElem<FP16>::from_f32(oacc[a][r] * inv_l);
       }
     }
     if (hi == 0) {
@@ -276,6 +275,7 @@ void fa_vl_fwd_kernel(const short* __restrict__ Q,
 // ---------------------------------------------------------------------------
 // 2. backward preprocess: delta = rowsum(dO * O)   [hq, total] fp32
 // ---------------------------------------------------------------------------
+template <bool FP16>
 __global__ void fa_vl_preprocess_kernel(const short* __restrict__ dO,
                                         const short* __restrict__ O,
                                         float* __restrict__ delta, int total,
@@ -289,8 +289,9 @@ __global__ void fa_vl_preprocess_kernel(const short* __restrict__ dO,
   const short* orow = O + row * D;
   float acc = 0.f;
   for (int i = lane * 2; i < D; i += WAVE * 2) {
-    acc += bf16_to_f32(dr[i]) * bf16_to_f32(orow[i]) +
-           bf16_to_f32(dr[i + 1]) * bf16_to_f32(orow[i + 1]);
+    acc += AttnElem<FP16>::to_f32(dr[i]) * AttnElem<FP16>::to_f32(orow[i]) +
+           AttnElem<FP16>::to_f32(dr[i + 1]) *
+               AttnElem<FP16>::to_f32(orow[i + 1]);
   }
   acc = wave_reduce_sum(acc);
   if (lane == 0) {
@@ -303,7 +304,7 @@ __global__ void fa_vl_preprocess_kernel(const short* __restrict__ dO,
 // ---------------------------------------------------------------------------
 // 3. dK/dV kernel
 // ---------------------------------------------------------------------------
-template <int D, bool CAUSAL>
+template <int D, bool CAUSAL, bool FP16>
 __global__ __launch_bounds__(256, 2)
 void fa_vl_dkv_kernel(const short* __restrict__ dOut,
                       const short* __restrict__ Q,
@@ -451,9 +452,8 @@ void fa_vl_dkv_kernel(const short* __restrict__ dOut,
           vbyte ^= (unsigned)((vrow & 7) << 4);
           bf16x8 vf = *reinterpret_cast<const bf16x8*>(
               reinterpret_cast<const char*>(v_lds) + vbyte);
-          s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfrag[t], s, 0, 0,
-                                                      0);
-          dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(df, vf, dp, 0, 0, 0);
+          s = AttnElem<FP16>::mfma(qf, kfrag[t], s);
+          dp = AttnElem<FP16>::mfma(df, vf, dp);
         }
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
@@ -487,17 +487,15 @@ void fa_vl_dkv_kernel(const short* __restrict__ dOut,
           attn_u32x4 ud_ = {dl_.x, dl_.y, dh_.x, dh_.y};                   \
           bf16x8 qbf = __builtin_bit_cast(bf16x8, uq_);                    \
           bf16x8 dob = __builtin_bit_cast(bf16x8, ud_);                    \
-          dvacc[a_] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(             \
-              pb, dob, dvacc[a_], 0, 0, 0);                                \
-          dkacc[a_] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(             \
-              dsb, qbf, dkacc[a_], 0, 0, 0);                               \
+          dvacc[a_] = AttnElem<FP16>::mfma(pb, dob, dvacc[a_]);            \
+          dkacc[a_] = AttnElem<FP16>::mfma(dsb, qbf, dkacc[a_]);           \
         }
 
 #pragma unroll
         for (int tp = 0; tp < 2; ++tp) {
           unsigned pfr[4], dsfr[4];
-          t12_pack_frag(s, tp, pfr);
-          t12_pack_frag(dp, tp, dsfr);
+          t12_pack_frag<AttnElem<FP16>>(s, tp, pfr);
+          t12_pack_frag<AttnElem<FP16>>(dp, tp, dsfr);
           bf16x8 pb = *reinterpret_cast<const bf16x8*>(pfr);
           bf16x8 dsb = *reinterpret_cast<const bf16x8*>(dsfr);
           if (tp == 0) {
@@ -530,8 +528,8 @@ void fa_vl_dkv_kernel(const short* __restrict__ dOut,
       const int d = a * 32 + col;
       if (key < total) {
         const long base = ((long)key * hk + kh) * D;
-        dK[base + d] = f32_to_bf16(dkacc[a][r]);
-        dV[base + d] = f32_to_bf16(dvacc[a][r]);
+        dK[base + d] = AttnElem<FP16>::from_f32(dkacc[a][r]);
+        dV[base + d] = AttnElem<FP16>::from_f32(dvacc[a][r]);
       }
     }
   }
@@ -540,7 +538,7 @@ void fa_vl_dkv_kernel(const short* __restrict__ dOut,
 // ---------------------------------------------------------------------------
 // 4. dQ kernel (forward structure, 128-key staged tiles)
 // ---------------------------------------------------------------------------
-template <int D, bool CAUSAL>
+template <int D, bool CAUSAL, bool FP16>
 __global__ __launch_bounds__(512, 2)
 void fa_vl_dq_kernel(const short* __restrict__ dOut,
                      const short* __restrict__ Q,
@@ -669,10 +667,8 @@ void fa_vl_dq_kernel(const short* __restrict__ dOut,
               reinterpret_cast<const char*>(k_lds) + byte);
           bf16x8 vf = *reinterpret_cast<const bf16x8*>(
               reinterpret_cast<const char*>(v_lds) + byte);
-          s[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfrag[t],
-                                                          s[kb], 0, 0, 0);
-          dp[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dofrag[t],
-                                                           dp[kb], 0, 0, 0);
+          s[kb] = AttnElem<FP16>::mfma(kf, qfrag[t], s[kb]);
+          dp[kb] = AttnElem<FP16>::mfma(vf, dofrag[t], dp[kb]);
         }
       }
       const bool lane_ok = row_ok && isfinite(lse_q);
@@ -703,13 +699,12 @@ void fa_vl_dq_kernel(const short* __restrict__ dOut,
                   ((a_) ^ 1) * 64));                                       \
         attn_u32x4 uk_ = {kl_.x, kl_.y, kh_.x, kh_.y};                     \
         bf16x8 ktb = __builtin_bit_cast(bf16x8, uk_);                      \
-        dqacc[a_] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(               \
-            dsb, ktb, dqacc[a_], 0, 0, 0);                                 \
+        dqacc[a_] = AttnElem<FP16>::mfma(dsb, ktb, dqacc[a_]);             \
       }
 #define VLQ_TR_TP(kb_, tp_)                                                \
       {                                                                    \
         unsigned dsfr[4];                                                  \
-        t12_pack_frag(dp[kb_], tp_, dsfr);                                 \
+        t12_pack_frag<AttnElem<FP16>>(dp[kb_], tp_, dsfr);                 \
         bf16x8 dsb = *reinterpret_cast<const bf16x8*>(dsfr);               \
         VLQ_TR_STEP(kb_, tp_, 0);                                          \
         VLQ_TR_STEP(kb_, tp_, 1);                                          \
@@ -737,7 +732,7 @@ void fa_vl_dq_kernel(const short* __restrict__ dOut,
       const int d = a * 32 + col;
       if (q_r < total) {
         const long obase = ((long)q_r * hq + h) * D;
-        dQ[obase + d] = f32_to_bf16(dqacc[a][r]);
+        dQ[obase + d] = AttnElem<FP16>::from_f32(dqacc[a][r]);
       }
     }
   }
@@ -754,8 +749,10 @@ std::vector<torch::Tensor> fa_varlen_forward(torch::Tensor q, torch::Tensor k,
                                              bool causal) {
   TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
               v.is_contiguous());
-  TORCH_CHECK(q.scalar_type() == torch::kBFloat16,
-              "fa_varlen_forward: bf16 only (CDNA4 MFMA path)");
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16 ||
+                  q.scalar_type() == torch::kHalf,
+              "fa_varlen_forward: bf16/fp16 only (CDNA4 MFMA path)");
+  const bool fp16 = q.scalar_type() == torch::kHalf;
   TORCH_CHECK(q.dim() == 3 && k.dim() == 3, "packed [total, h, d] expected");
   const int total = q.size(0), hq = q.size(1), D = q.size(2);
   const int hk = k.size(1);
@@ -772,7 +769,16 @@ std::vector<torch::Tensor> fa_varlen_forward(torch::Tensor q, torch::Tensor k,
   dim3 grid((total + 255) / 256, hq), block(512);
 
 #define LAUNCH_F(DD, C)                                                      \
-  hipLaunchKernelGGL((fa_vl_fwd_kernel<DD, C>), grid, block,                 \
+  if (fp16)                                                                  \
+    hipLaunchKernelGGL((fa_vl_fwd_kernel<DD, C, true>), grid, block,         \
+                       4 * 64 * DD * 2, stream, (const short*)q.data_ptr(),  \
+                       (const short*)k.data_ptr(),                           \
+                       (const short*)v.data_ptr(), (short*)o.data_ptr(),     \
+                       lse.data_ptr<float>(),                                \
+                       (const int2*)bounds.data_ptr<int>(), total, hq, hk,   \
+                       (float)softmax_scale);                                \
+  else                                                                       \
+    hipLaunchKernelGGL((fa_vl_fwd_kernel<DD, C, false>), grid, block,        \
                      4 * 64 * DD * 2, stream, (const short*)q.data_ptr(),    \
                      (const short*)k.data_ptr(),                             \
                      (const short*)v.data_ptr(), (short*)o.data_ptr(),       \
@@ -792,7 +798,9 @@ std::vector<torch::Tensor> fa_varlen_backward(
     double softmax_scale, bool causal) {
   TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
               v.is_contiguous() && out.is_contiguous());
-  TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16 ||
+              q.scalar_type() == torch::kHalf);
+  const bool fp16 = q.scalar_type() == torch::kHalf;
   dout = dout.contiguous();
   const int total = q.size(0), hq = q.size(1), D = q.size(2);
   const int hk = k.size(1);
@@ -808,10 +816,16 @@ std::vector<torch::Tensor> fa_varlen_backward(
     const long rows = (long)total * hq;
     const int rpb = 256 / WAVE;
     dim3 g((rows + rpb - 1) / rpb), bl(256);
-    hipLaunchKernelGGL(fa_vl_preprocess_kernel, g, bl, 0, stream,
-                       (const short*)dout.data_ptr(),
-                       (const short*)out.data_ptr(),
-                       delta.data_ptr<float>(), total, hq, D);
+    if (fp16)
+      hipLaunchKernelGGL(fa_vl_preprocess_kernel<true>, g, bl, 0, stream,
+                         (const short*)dout.data_ptr(),
+                         (const short*)out.data_ptr(),
+                         delta.data_ptr<float>(), total, hq, D);
+    else
+      hipLaunchKernelGGL(fa_vl_preprocess_kernel<false>, g, bl, 0, stream,
+                         (const short*)dout.data_ptr(),
+                         (const short*)out.data_ptr(),
+                         delta.data_ptr<float>(), total, hq, D);
   }
 
   dim3 gkv((total + 127) / 128, hk), bkv(256);
@@ -822,7 +836,20 @@ std::vector<torch::Tensor> fa_varlen_backward(
     const int lds_kv =                                                       \
         (2 * 32 * DD + 128 * DD) * 2 + 2 * 32 * 4 + 32 * 8;                  \
     const int lds_q = 2 * 128 * DD * 2;                                      \
-    hipLaunchKernelGGL((fa_vl_dkv_kernel<DD, C>), gkv, bkv, lds_kv, stream,  \
+    if (fp16)                                                              \
+      hipLaunchKernelGGL((fa_vl_dkv_kernel<DD, C, true>), gkv, bkv, lds_kv, \
+                       stream,                                              \
+                       (const short*)dout.data_ptr(),                       \
+                       (const short*)q.data_ptr(),                          \
+                       (const short*)k.data_ptr(),                          \
+                       (const short*)v.data_ptr(), lse.data_ptr<float>(),   \
+                       delta.data_ptr<float>(), (short*)dk.data_ptr(),      \
+                       (short*)dv.data_ptr(),                               \
+                       (const int2*)bounds.data_ptr<int>(), total, hq, hk,  \
+                       (float)softmax_scale);                               \
+    else                                                                    \
+      hipLaunchKernelGGL((fa_vl_dkv_kernel<DD, C, false>), gkv, bkv,        \
+                       lds_kv, stream,                                      \
                        (const short*)dout.data_ptr(),                        \
                        (const short*)q.data_ptr(),                           \
                        (const short*)k.data_ptr(),                           \
@@ -831,7 +858,19 @@ std::vector<torch::Tensor> fa_varlen_backward(
                        (short*)dv.data_ptr(),                                \
                        (const int2*)bounds.data_ptr<int>(), total, hq, hk,   \
                        (float)softmax_scale);                                \
-    hipLaunchKernelGGL((fa_vl_dq_kernel<DD, C>), gq, bq, lds_q, stream,      \
+    if (fp16)                                                              \
+      hipLaunchKernelGGL((fa_vl_dq_kernel<DD, C, true>), gq, bq, lds_q,     \
+                       stream,                                              \
+                       (const short*)dout.data_ptr(),                       \
+                       (const short*)q.data_ptr(),                          \
+                       (const short*)k.data_ptr(),                          \
+                       (const short*)v.data_ptr(), lse.data_ptr<float>(),   \
+                       delta.data_ptr<float>(), (short*)dq.data_ptr(),      \
+                       (const int2*)bounds.data_ptr<int>(), total, hq, hk,  \
+                       (float)softmax_scale);                               \
+    else                                                                    \
+      hipLaunchKernelGGL((fa_vl_dq_kernel<DD, C, false>), gq, bq, lds_q,    \
+                       stream,                                              \
                        (const short*)dout.data_ptr(),                        \
                        (const short*)q.data_ptr(),                           \
                        (const short*)k.data_ptr(),                           \

@@ -87,13 +87,14 @@ def _pad_target(d: int):
 
 
 def _kernel_ext(q):
-    """The CDNA4 kernels cover bf16 with head_dim 64/128 natively and any
-    d <= 128 via zero padding; d > 128 runs the fp32 composite (warned
-    once per dim — the flagship models are all head_dim 128)."""
+    """The CDNA4 kernels cover bf16 AND fp16 (AttnElem-templated MFMA
+    ladder) with head_dim 64/128 natively and any d <= 128 via zero
+    padding; d > 128 runs the fp32 composite (warned once per dim — the
+    flagship models are all head_dim 128)."""
     ext = dispatch(q)
     if ext is None:
         return None
-    if q.dtype != torch.bfloat16:
+    if q.dtype not in (torch.bfloat16, torch.float16):
         return None
     d = q.shape[-1]
     if _pad_target(d) == -1:
@@ -325,23 +326,10 @@ def _ref_fa_backward(dout, q, k, v, out, lse, softmax_scale, causal, window,
 # user wrappers (reference-compatible names)
 # ---------------------------------------------------------------------------
 
-_warned_fp16 = [False]
-
-
 def _cast_fp16(q, k, v):
-    """fp16 inputs on GPU run the bf16 MFMA kernels through an explicit
-    (warned, autograd-tracked) cast — never the fp32 composite, which would
-    materialize [b,h,s,s] scores (reference accepts fp16 natively,
-    ops/flash_attn.py:324-325). Returns (q, k, v, needs_cast_back)."""
-    if q.is_cuda and q.dtype == torch.float16:
-        if not _warned_fp16[0]:
-            _warned_fp16[0] = True
-            from ..utils.logger import logger
-            logger.warning(
-                "flash-attention fp16 inputs: computing in bf16 on the "
-                "CDNA4 MFMA kernels (output cast back to fp16)")
-        return q.to(torch.bfloat16), k.to(torch.bfloat16), \
-            v.to(torch.bfloat16), True
+    """fp16 is now a NATIVE kernel dtype (v_mfma_f32_32x32x16_f16 via the
+    AttnElem traits) — this shim is a passthrough kept for API stability
+    (reference accepts fp16, ops/flash_attn.py:324-325)."""
     return q, k, v, False
 
 
