@@ -108,7 +108,8 @@ def main():
         cfg.memory.gc_cls = {"LlamaDecoderLayer", "Qwen2DecoderLayer"}
         cfg.memory.gc_selective_attn = not args.no_gc_selective
         gc_cnt = args.gc_cnt
-        if gc_cnt is None and args.model == "llama-2-7b" \
+        if gc_cnt is None \
+                and args.model in ("llama-2-7b", "llama-3-8b", "qwen2-7b") \
                 and args.mode == "fsdp" \
                 and args.batch_size * args.seq_len <= 32768:
             # 288 GB HBM3E rarely needs every layer checkpointed: 8 of 32
