@@ -116,3 +116,15 @@ DEVINLINE bool attn_dropout_keep(unsigned long long seed,
   x ^= x >> 31;
   return ((unsigned)x & 0xFFFFFFu) >= threshold24;
 }
+
+// host-side dtype dispatch: runs BODY with constexpr kFP16 bound
+#define FP16_SWITCH(COND, ...)                         \
+  do {                                                 \
+    if (COND) {                                        \
+      constexpr bool kFP16 = true;                     \
+      __VA_ARGS__;                                     \
+    } else {                                           \
+      constexpr bool kFP16 = false;                    \
+      __VA_ARGS__;                                     \
+    }                                                  \
+  } while (0)

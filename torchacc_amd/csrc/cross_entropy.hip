@@ -7,7 +7,9 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include "common.h"
+#include "attn_common.h"
 
+template <bool FP16>
 __global__ void ce_fwd_kernel(const short* __restrict__ logits,
                               const long* __restrict__ target,
                               float* __restrict__ lse,
@@ -24,10 +26,10 @@ __global__ void ce_fwd_kernel(const short* __restrict__ logits,
       if (i + 8 <= V) {
         s16x8 v = *reinterpret_cast<const s16x8*>(lr + i);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) vals[j] = bf16_to_f32(v[j]);
+        for (int j = 0; j < 8; ++j) vals[j] = AttnElem<FP16>::to_f32(v[j]);
       } else {
         for (int j = 0; j < 8; ++j)
-          vals[j] = (i + j < V) ? bf16_to_f32(lr[i + j]) : -INFINITY;
+          vals[j] = (i + j < V) ? AttnElem<FP16>::to_f32(lr[i + j]) : -INFINITY;
       }
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -70,7 +72,7 @@ __global__ void ce_fwd_kernel(const short* __restrict__ logits,
       lse[row] = l;
       long t = target[row];
       if (t != ignore_index) {
-        float picked = bf16_to_f32(lr[t]);
+        float picked = AttnElem<FP16>::to_f32(lr[t]);
         atomicAdd(loss_sum, l - picked);
         atomicAdd(nvalid, 1);
       }
@@ -80,6 +82,7 @@ __global__ void ce_fwd_kernel(const short* __restrict__ logits,
   (void)red;
 }
 
+template <bool FP16>
 __global__ void ce_bwd_kernel(const short* __restrict__ logits,
                               const long* __restrict__ target,
                               const float* __restrict__ lse,
@@ -103,16 +106,16 @@ __global__ void ce_bwd_kernel(const short* __restrict__ logits,
       s16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float soft = __expf(bf16_to_f32(v[j]) - l);
+        float soft = __expf(AttnElem<FP16>::to_f32(v[j]) - l);
         if (valid && (long)(i + j) == t) soft -= 1.f;
-        o[j] = f32_to_bf16(valid ? soft * sc : 0.f);
+        o[j] = AttnElem<FP16>::from_f32(valid ? soft * sc : 0.f);
       }
       *reinterpret_cast<s16x8*>(dr + i) = o;
     } else {
       for (int j = 0; i + j < V; ++j) {
-        float soft = __expf(bf16_to_f32(lr[i + j]) - l);
+        float soft = __expf(AttnElem<FP16>::to_f32(lr[i + j]) - l);
         if (valid && (long)(i + j) == t) soft -= 1.f;
-        dr[i + j] = f32_to_bf16(valid ? soft * sc : 0.f);
+        dr[i + j] = AttnElem<FP16>::from_f32(valid ? soft * sc : 0.f);
       }
     }
   }
@@ -122,7 +125,8 @@ std::vector<torch::Tensor> cross_entropy_forward(torch::Tensor logits,
                                                  torch::Tensor target,
                                                  long ignore_index) {
   TORCH_CHECK(logits.is_cuda() && logits.dim() == 2);
-  TORCH_CHECK(logits.scalar_type() == torch::kBFloat16,
+  const bool f16 = logits.scalar_type() == torch::kHalf;
+  TORCH_CHECK(logits.scalar_type() == torch::kBFloat16 || f16,
               "cross_entropy: bf16 logits only on GPU");
   TORCH_CHECK(target.scalar_type() == torch::kLong);
   const int rows = logits.size(0), V = logits.size(1);
@@ -130,11 +134,13 @@ std::vector<torch::Tensor> cross_entropy_forward(torch::Tensor logits,
   auto loss = torch::zeros({}, logits.options().dtype(torch::kFloat32));
   auto nvalid = torch::zeros({}, logits.options().dtype(torch::kInt32));
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(ce_fwd_kernel, dim3(std::min(rows, 2048)), dim3(256), 0,
-                     stream, (const short*)logits.data_ptr(),
-                     target.data_ptr<long>(), lse.data_ptr<float>(),
-                     loss.data_ptr<float>(), nvalid.data_ptr<int>(), rows, V,
-                     ignore_index);
+  FP16_SWITCH(f16,
+      hipLaunchKernelGGL((ce_fwd_kernel<kFP16>),
+                         dim3(std::min(rows, 2048)), dim3(256), 0, stream,
+                         (const short*)logits.data_ptr(),
+                         target.data_ptr<long>(), lse.data_ptr<float>(),
+                         loss.data_ptr<float>(), nvalid.data_ptr<int>(),
+                         rows, V, ignore_index));
   HIP_CHECK_LAST();
   return {loss, nvalid, lse};
 }
@@ -148,11 +154,14 @@ torch::Tensor cross_entropy_backward(torch::Tensor logits,
   auto scale_f = scale.to(torch::kFloat32);
   const long work = (long)rows * ((V + 7) / 8);
   int grid = (int)std::min<long>((work + 255) / 256, 2048);
-  hipLaunchKernelGGL(ce_bwd_kernel, dim3(grid), dim3(256), 0, stream,
-                     (const short*)logits.data_ptr(),
-                     target.data_ptr<long>(), lse.data_ptr<float>(),
-                     scale_f.data_ptr<float>(), (short*)dlogits.data_ptr(),
-                     rows, V, ignore_index);
+  const bool f16b = logits.scalar_type() == torch::kHalf;
+  FP16_SWITCH(f16b,
+      hipLaunchKernelGGL((ce_bwd_kernel<kFP16>), dim3(grid), dim3(256), 0,
+                         stream, (const short*)logits.data_ptr(),
+                         target.data_ptr<long>(), lse.data_ptr<float>(),
+                         scale_f.data_ptr<float>(),
+                         (short*)dlogits.data_ptr(), rows, V,
+                         ignore_index));
   HIP_CHECK_LAST();
   return dlogits;
 }

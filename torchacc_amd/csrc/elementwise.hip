@@ -6,6 +6,7 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include "common.h"
+#include "attn_common.h"
 
 // ---------------------------------------------------------------------------
 // RMSNorm
@@ -16,6 +17,7 @@ template <typename T>
 struct VecIO;
 
 // bf16 path: 8 elements per 16B
+template <bool FP16>
 __global__ void rmsnorm_fwd_kernel(const short* __restrict__ x,
                                    const short* __restrict__ w,
                                    short* __restrict__ y,
@@ -30,7 +32,7 @@ __global__ void rmsnorm_fwd_kernel(const short* __restrict__ x,
       s16x8 v = *reinterpret_cast<const s16x8*>(xr + i);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float f = bf16_to_f32(v[j]);
+        float f = AttnElem<FP16>::to_f32(v[j]);
         ss += f * f;
       }
     }
@@ -43,7 +45,7 @@ __global__ void rmsnorm_fwd_kernel(const short* __restrict__ x,
       s16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        o[j] = f32_to_bf16(bf16_to_f32(v[j]) * r * bf16_to_f32(wv[j]));
+        o[j] = AttnElem<FP16>::from_f32(AttnElem<FP16>::to_f32(v[j]) * r * AttnElem<FP16>::to_f32(wv[j]));
       *reinterpret_cast<s16x8*>(yr + i) = o;
     }
   }
@@ -51,6 +53,7 @@ __global__ void rmsnorm_fwd_kernel(const short* __restrict__ x,
 
 // backward: dx per row + dw accumulated in LDS across this block's rows,
 // then one fp32 atomicAdd per element (guide G12: partial-reduce first).
+template <bool FP16>
 __global__ void rmsnorm_bwd_kernel(const short* __restrict__ dy,
                                    const short* __restrict__ x,
                                    const short* __restrict__ w,
@@ -76,8 +79,8 @@ __global__ void rmsnorm_bwd_kernel(const short* __restrict__ dy,
       s16x8 wv = *reinterpret_cast<const s16x8*>(w + i);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float xh = bf16_to_f32(xv[j]) * r;
-        dot += bf16_to_f32(wv[j]) * bf16_to_f32(dv[j]) * xh;
+        float xh = AttnElem<FP16>::to_f32(xv[j]) * r;
+        dot += AttnElem<FP16>::to_f32(wv[j]) * AttnElem<FP16>::to_f32(dv[j]) * xh;
       }
     }
     dot = block_reduce_sum<4>(dot, red) / H;
@@ -89,9 +92,9 @@ __global__ void rmsnorm_bwd_kernel(const short* __restrict__ dy,
       s16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float xh = bf16_to_f32(xv[j]) * r;
-        float dyf = bf16_to_f32(dv[j]);
-        o[j] = f32_to_bf16(r * (bf16_to_f32(wv[j]) * dyf - xh * dot) +
+        float xh = AttnElem<FP16>::to_f32(xv[j]) * r;
+        float dyf = AttnElem<FP16>::to_f32(dv[j]);
+        o[j] = AttnElem<FP16>::from_f32(r * (AttnElem<FP16>::to_f32(wv[j]) * dyf - xh * dot) +
                            0.f);
         dw_acc[i + j] += dyf * xh;  // each thread owns cols i..i+7
       }
@@ -108,7 +111,7 @@ __global__ void rmsnorm_bwd_kernel(const short* __restrict__ dy,
 // never re-reads the row, and dw-style reductions stay in registers.
 // Covers Llama H=4096/8192 (ITERS 2/4); other widths take the generic
 // kernels below.
-template <int ITERS>
+template <int ITERS, bool FP16>
 __global__ void add_rmsnorm_fwd_fast(const short* __restrict__ x,
                                      const short* __restrict__ resid_in,
                                      const short* __restrict__ w,
@@ -139,18 +142,18 @@ __global__ void add_rmsnorm_fwd_fast(const short* __restrict__ x,
         s16x8 rv = *reinterpret_cast<const s16x8*>(rr + i);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          float f = bf16_to_f32(v[j]) + bf16_to_f32(rv[j]);
+          float f = AttnElem<FP16>::to_f32(v[j]) + AttnElem<FP16>::to_f32(rv[j]);
           // match the generic kernel: resid_out stores the bf16 ROUNDING
           // of the sum, and the norm pass reads that rounded value
-          o[j] = f32_to_bf16(f);
-          float fr = bf16_to_f32(o[j]);
+          o[j] = AttnElem<FP16>::from_f32(f);
+          float fr = AttnElem<FP16>::to_f32(o[j]);
           rc[t][j / 4][j % 4] = fr;
           ss += fr * fr;
         }
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          float f = bf16_to_f32(v[j]);
+          float f = AttnElem<FP16>::to_f32(v[j]);
           o[j] = v[j];
           rc[t][j / 4][j % 4] = f;
           ss += f * f;
@@ -167,13 +170,13 @@ __global__ void add_rmsnorm_fwd_fast(const short* __restrict__ x,
       s16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        o[j] = f32_to_bf16(rc[t][j / 4][j % 4] * r * bf16_to_f32(wv[t][j]));
+        o[j] = AttnElem<FP16>::from_f32(rc[t][j / 4][j % 4] * r * AttnElem<FP16>::to_f32(wv[t][j]));
       *reinterpret_cast<s16x8*>(yr + i) = o;
     }
   }
 }
 
-template <int ITERS>
+template <int ITERS, bool FP16>
 __global__ void add_rmsnorm_bwd_fast(const short* __restrict__ dy,
                                      const short* __restrict__ dresid,
                                      const short* __restrict__ r_saved,
@@ -208,8 +211,8 @@ __global__ void add_rmsnorm_bwd_fast(const short* __restrict__ dy,
       dc[t] = *reinterpret_cast<const s16x8*>(dyr + i);
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        dot += bf16_to_f32(wv[t][j]) * bf16_to_f32(dc[t][j]) *
-               (bf16_to_f32(xc[t][j]) * rinv);
+        dot += AttnElem<FP16>::to_f32(wv[t][j]) * AttnElem<FP16>::to_f32(dc[t][j]) *
+               (AttnElem<FP16>::to_f32(xc[t][j]) * rinv);
     }
     dot = block_reduce_sum<4>(dot, red) / H;
 #pragma unroll
@@ -220,11 +223,11 @@ __global__ void add_rmsnorm_bwd_fast(const short* __restrict__ dy,
       if (has_dresid) drv = *reinterpret_cast<const s16x8*>(drr + i);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float xh = bf16_to_f32(xc[t][j]) * rinv;
-        float dyf = bf16_to_f32(dc[t][j]);
-        float g = rinv * (bf16_to_f32(wv[t][j]) * dyf - xh * dot);
-        if (has_dresid) g += bf16_to_f32(drv[j]);
-        o[j] = f32_to_bf16(g);
+        float xh = AttnElem<FP16>::to_f32(xc[t][j]) * rinv;
+        float dyf = AttnElem<FP16>::to_f32(dc[t][j]);
+        float g = rinv * (AttnElem<FP16>::to_f32(wv[t][j]) * dyf - xh * dot);
+        if (has_dresid) g += AttnElem<FP16>::to_f32(drv[j]);
+        o[j] = AttnElem<FP16>::from_f32(g);
         dwacc[t][j / 4][j % 4] += dyf * xh;
       }
       *reinterpret_cast<s16x8*>(dxr + i) = o;
@@ -242,6 +245,7 @@ __global__ void add_rmsnorm_bwd_fast(const short* __restrict__ dy,
 // fused residual add + RMSNorm: r = x + resid_in; y = rmsnorm(r) * w.
 // Saves one full read+write of the residual stream per call vs separate
 // add and norm kernels (and the separate add's backward elementwise).
+template <bool FP16>
 __global__ void add_rmsnorm_fwd_kernel(const short* __restrict__ x,
                                        const short* __restrict__ resid_in,
                                        const short* __restrict__ w,
@@ -263,14 +267,14 @@ __global__ void add_rmsnorm_fwd_kernel(const short* __restrict__ x,
         s16x8 rv = *reinterpret_cast<const s16x8*>(rr + i);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          float f = bf16_to_f32(v[j]) + bf16_to_f32(rv[j]);
-          o[j] = f32_to_bf16(f);
+          float f = AttnElem<FP16>::to_f32(v[j]) + AttnElem<FP16>::to_f32(rv[j]);
+          o[j] = AttnElem<FP16>::from_f32(f);
           ss += f * f;
         }
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          float f = bf16_to_f32(v[j]);
+          float f = AttnElem<FP16>::to_f32(v[j]);
           o[j] = v[j];
           ss += f * f;
         }
@@ -286,7 +290,7 @@ __global__ void add_rmsnorm_fwd_kernel(const short* __restrict__ x,
       s16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        o[j] = f32_to_bf16(bf16_to_f32(v[j]) * r * bf16_to_f32(wv[j]));
+        o[j] = AttnElem<FP16>::from_f32(AttnElem<FP16>::to_f32(v[j]) * r * AttnElem<FP16>::to_f32(wv[j]));
       *reinterpret_cast<s16x8*>(yr + i) = o;
     }
   }
@@ -294,6 +298,7 @@ __global__ void add_rmsnorm_fwd_kernel(const short* __restrict__ x,
 
 // backward: dx = rmsnorm_dx(dy) + dresid (grad into the residual stream
 // from downstream); dw accumulated as in rmsnorm_bwd.
+template <bool FP16>
 __global__ void add_rmsnorm_bwd_kernel(const short* __restrict__ dy,
                                        const short* __restrict__ dresid,
                                        const short* __restrict__ r_saved,
@@ -320,8 +325,8 @@ __global__ void add_rmsnorm_bwd_kernel(const short* __restrict__ dy,
       s16x8 wv = *reinterpret_cast<const s16x8*>(w + i);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float xh = bf16_to_f32(xv[j]) * rinv;
-        dot += bf16_to_f32(wv[j]) * bf16_to_f32(dv[j]) * xh;
+        float xh = AttnElem<FP16>::to_f32(xv[j]) * rinv;
+        dot += AttnElem<FP16>::to_f32(wv[j]) * AttnElem<FP16>::to_f32(dv[j]) * xh;
       }
     }
     dot = block_reduce_sum<4>(dot, red) / H;
@@ -332,14 +337,14 @@ __global__ void add_rmsnorm_bwd_kernel(const short* __restrict__ dy,
       s16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float xh = bf16_to_f32(xv[j]) * rinv;
-        float dyf = bf16_to_f32(dv[j]);
-        float g = rinv * (bf16_to_f32(wv[j]) * dyf - xh * dot);
+        float xh = AttnElem<FP16>::to_f32(xv[j]) * rinv;
+        float dyf = AttnElem<FP16>::to_f32(dv[j]);
+        float g = rinv * (AttnElem<FP16>::to_f32(wv[j]) * dyf - xh * dot);
         if (has_dresid) {
           s16x8 drv = *reinterpret_cast<const s16x8*>(drr + i);
-          g += bf16_to_f32(drv[j]);
+          g += AttnElem<FP16>::to_f32(drv[j]);
         }
-        o[j] = f32_to_bf16(g);
+        o[j] = AttnElem<FP16>::from_f32(g);
         dw_acc[i + j] += dyf * xh;
       }
       *reinterpret_cast<s16x8*>(dxr + i) = o;
@@ -376,6 +381,7 @@ __global__ void rmsnorm_fwd_kernel_f32(const float* __restrict__ x,
 // x [b, s, h, d]; cos/sin [s, d/2] fp32. Each thread: 8 first-half elements
 // + 8 matching second-half elements of one (b,s,h) row.
 
+template <bool FP16>
 __global__ void rope_kernel(const short* __restrict__ x,
                             short* __restrict__ out,
                             const float* __restrict__ cosp,
@@ -404,10 +410,10 @@ __global__ void rope_kernel(const short* __restrict__ x,
     for (int j = 0; j < 8; ++j) {
       float cj = j < 4 ? c0[j] : c1[j - 4];
       float sj = j < 4 ? s0[j] : s1[j - 4];
-      float a = bf16_to_f32(x1[j]);
-      float b = bf16_to_f32(x2[j]);
-      o1[j] = f32_to_bf16(a * cj - b * sj);
-      o2[j] = f32_to_bf16(b * cj + a * sj);
+      float a = AttnElem<FP16>::to_f32(x1[j]);
+      float b = AttnElem<FP16>::to_f32(x2[j]);
+      o1[j] = AttnElem<FP16>::from_f32(a * cj - b * sj);
+      o2[j] = AttnElem<FP16>::from_f32(b * cj + a * sj);
     }
     *reinterpret_cast<s16x8*>(orow) = o1;
     *reinterpret_cast<s16x8*>(orow + d2) = o2;
@@ -421,6 +427,7 @@ __global__ void rope_kernel(const short* __restrict__ x,
 
 // 2x-wide variants: two s16x8 (32 B) per thread per trip doubles the
 // loads in flight per wave — the 1x versions measured ~60% of HBM peak.
+template <bool FP16>
 __global__ void swiglu_fwd_kernel_v2(const short* __restrict__ g,
                                      const short* __restrict__ u,
                                      short* __restrict__ y, long n16) {
@@ -434,18 +441,19 @@ __global__ void swiglu_fwd_kernel_v2(const short* __restrict__ g,
     s16x8 o0, o1;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      float gf0 = bf16_to_f32(gv0[j]);
-      float gf1 = bf16_to_f32(gv1[j]);
+      float gf0 = AttnElem<FP16>::to_f32(gv0[j]);
+      float gf1 = AttnElem<FP16>::to_f32(gv1[j]);
       float s0 = 1.f / (1.f + __expf(-gf0));
       float s1 = 1.f / (1.f + __expf(-gf1));
-      o0[j] = f32_to_bf16(gf0 * s0 * bf16_to_f32(uv0[j]));
-      o1[j] = f32_to_bf16(gf1 * s1 * bf16_to_f32(uv1[j]));
+      o0[j] = AttnElem<FP16>::from_f32(gf0 * s0 * AttnElem<FP16>::to_f32(uv0[j]));
+      o1[j] = AttnElem<FP16>::from_f32(gf1 * s1 * AttnElem<FP16>::to_f32(uv1[j]));
     }
     *reinterpret_cast<s16x8*>(y + base) = o0;
     *reinterpret_cast<s16x8*>(y + base + 8) = o1;
   }
 }
 
+template <bool FP16>
 __global__ void swiglu_bwd_kernel_v2(const short* __restrict__ dy,
                                      const short* __restrict__ g,
                                      const short* __restrict__ u,
@@ -463,18 +471,18 @@ __global__ void swiglu_bwd_kernel_v2(const short* __restrict__ dy,
     s16x8 og0, og1, ou0, ou1;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      float gf0 = bf16_to_f32(gv0[j]);
-      float gf1 = bf16_to_f32(gv1[j]);
-      float d0 = bf16_to_f32(dv0[j]);
-      float d1 = bf16_to_f32(dv1[j]);
+      float gf0 = AttnElem<FP16>::to_f32(gv0[j]);
+      float gf1 = AttnElem<FP16>::to_f32(gv1[j]);
+      float d0 = AttnElem<FP16>::to_f32(dv0[j]);
+      float d1 = AttnElem<FP16>::to_f32(dv1[j]);
       float s0 = 1.f / (1.f + __expf(-gf0));
       float s1 = 1.f / (1.f + __expf(-gf1));
-      og0[j] = f32_to_bf16(d0 * bf16_to_f32(uv0[j]) *
+      og0[j] = AttnElem<FP16>::from_f32(d0 * AttnElem<FP16>::to_f32(uv0[j]) *
                            (s0 * (1.f + gf0 * (1.f - s0))));
-      og1[j] = f32_to_bf16(d1 * bf16_to_f32(uv1[j]) *
+      og1[j] = AttnElem<FP16>::from_f32(d1 * AttnElem<FP16>::to_f32(uv1[j]) *
                            (s1 * (1.f + gf1 * (1.f - s1))));
-      ou0[j] = f32_to_bf16(d0 * gf0 * s0);
-      ou1[j] = f32_to_bf16(d1 * gf1 * s1);
+      ou0[j] = AttnElem<FP16>::from_f32(d0 * gf0 * s0);
+      ou1[j] = AttnElem<FP16>::from_f32(d1 * gf1 * s1);
     }
     *reinterpret_cast<s16x8*>(dg + base) = og0;
     *reinterpret_cast<s16x8*>(dg + base + 8) = og1;
@@ -483,6 +491,7 @@ __global__ void swiglu_bwd_kernel_v2(const short* __restrict__ dy,
   }
 }
 
+template <bool FP16>
 __global__ void swiglu_fwd_kernel(const short* __restrict__ g,
                                   const short* __restrict__ u,
                                   short* __restrict__ y, long n8) {
@@ -493,14 +502,15 @@ __global__ void swiglu_fwd_kernel(const short* __restrict__ g,
     s16x8 o;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      float gf = bf16_to_f32(gv[j]);
+      float gf = AttnElem<FP16>::to_f32(gv[j]);
       float sig = 1.f / (1.f + __expf(-gf));
-      o[j] = f32_to_bf16(gf * sig * bf16_to_f32(uv[j]));
+      o[j] = AttnElem<FP16>::from_f32(gf * sig * AttnElem<FP16>::to_f32(uv[j]));
     }
     *reinterpret_cast<s16x8*>(y + idx * 8) = o;
   }
 }
 
+template <bool FP16>
 __global__ void swiglu_bwd_kernel(const short* __restrict__ dy,
                                   const short* __restrict__ g,
                                   const short* __restrict__ u,
@@ -514,13 +524,13 @@ __global__ void swiglu_bwd_kernel(const short* __restrict__ dy,
     s16x8 og, ou;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      float gf = bf16_to_f32(gv[j]);
-      float uf = bf16_to_f32(uv[j]);
-      float d = bf16_to_f32(dv[j]);
+      float gf = AttnElem<FP16>::to_f32(gv[j]);
+      float uf = AttnElem<FP16>::to_f32(uv[j]);
+      float d = AttnElem<FP16>::to_f32(dv[j]);
       float sig = 1.f / (1.f + __expf(-gf));
       float silu = gf * sig;
-      og[j] = f32_to_bf16(d * uf * (sig * (1.f + gf * (1.f - sig))));
-      ou[j] = f32_to_bf16(d * silu);
+      og[j] = AttnElem<FP16>::from_f32(d * uf * (sig * (1.f + gf * (1.f - sig))));
+      ou[j] = AttnElem<FP16>::from_f32(d * silu);
     }
     *reinterpret_cast<s16x8*>(dg + idx * 8) = og;
     *reinterpret_cast<s16x8*>(du + idx * 8) = ou;
@@ -545,13 +555,17 @@ std::vector<torch::Tensor> rmsnorm_forward(torch::Tensor x, torch::Tensor w,
   auto inv_rms = torch::empty({rows}, x.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
   const int grid = grid_for(rows, 1);
-  if (x.scalar_type() == torch::kBFloat16) {
+  if (x.scalar_type() == torch::kBFloat16 ||
+      x.scalar_type() == torch::kHalf) {
     TORCH_CHECK(H % 8 == 0, "hidden size must be a multiple of 8");
-    hipLaunchKernelGGL(rmsnorm_fwd_kernel, dim3(std::min<long>(rows, 2048)),
-                       dim3(256), 0, stream,
-                       (const short*)x.data_ptr(), (const short*)w.data_ptr(),
-                       (short*)y.data_ptr(), inv_rms.data_ptr<float>(),
-                       (int)rows, H, (float)eps);
+    FP16_SWITCH(x.scalar_type() == torch::kHalf,
+        hipLaunchKernelGGL((rmsnorm_fwd_kernel<kFP16>),
+                           dim3(std::min<long>(rows, 2048)),
+                           dim3(256), 0, stream,
+                           (const short*)x.data_ptr(),
+                           (const short*)w.data_ptr(),
+                           (short*)y.data_ptr(), inv_rms.data_ptr<float>(),
+                           (int)rows, H, (float)eps));
   } else if (x.scalar_type() == torch::kFloat32) {
     hipLaunchKernelGGL(rmsnorm_fwd_kernel_f32,
                        dim3(std::min<long>(rows, 2048)), dim3(256), 0, stream,
@@ -568,19 +582,25 @@ std::vector<torch::Tensor> rmsnorm_forward(torch::Tensor x, torch::Tensor w,
 std::vector<torch::Tensor> rmsnorm_backward(torch::Tensor dy, torch::Tensor x,
                                             torch::Tensor w,
                                             torch::Tensor inv_rms) {
-  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16,
-              "rmsnorm backward: bf16 only on GPU");
+  TORCH_CHECK(x.is_cuda() && (x.scalar_type() == torch::kBFloat16 ||
+                              x.scalar_type() == torch::kHalf),
+              "rmsnorm backward: bf16/fp16 only on GPU");
+  const bool f16 = x.scalar_type() == torch::kHalf;
   const int H = x.size(-1);
   const long rows = x.numel() / H;
   auto dx = torch::empty_like(x);
   auto dw32 = torch::zeros({H}, x.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
   const int lds = (H + 16) * sizeof(float);
-  hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(std::min<long>(rows, 512)),
-                     dim3(256), lds, stream, (const short*)dy.data_ptr(),
-                     (const short*)x.data_ptr(), (const short*)w.data_ptr(),
-                     inv_rms.data_ptr<float>(), (short*)dx.data_ptr(),
-                     dw32.data_ptr<float>(), (int)rows, H);
+  FP16_SWITCH(f16,
+      hipLaunchKernelGGL((rmsnorm_bwd_kernel<kFP16>),
+                         dim3(std::min<long>(rows, 512)),
+                         dim3(256), lds, stream,
+                         (const short*)dy.data_ptr(),
+                         (const short*)x.data_ptr(),
+                         (const short*)w.data_ptr(),
+                         inv_rms.data_ptr<float>(), (short*)dx.data_ptr(),
+                         dw32.data_ptr<float>(), (int)rows, H));
   HIP_CHECK_LAST();
   return {dx, dw32.to(w.scalar_type())};
 }
@@ -588,7 +608,10 @@ std::vector<torch::Tensor> rmsnorm_backward(torch::Tensor dy, torch::Tensor x,
 std::vector<torch::Tensor> rope_forward(torch::Tensor q, torch::Tensor k,
                                         torch::Tensor cos, torch::Tensor sin) {
   TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous());
-  TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "rope: bf16 only on GPU");
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16 ||
+                  q.scalar_type() == torch::kHalf,
+              "rope: bf16/fp16 only on GPU");
+  const bool f16 = q.scalar_type() == torch::kHalf;
   TORCH_CHECK(cos.scalar_type() == torch::kFloat32);
   const int D = q.size(-1);
   TORCH_CHECK(D % 16 == 0, "rope: head_dim must be a multiple of 16");
@@ -601,41 +624,53 @@ std::vector<torch::Tensor> rope_forward(torch::Tensor q, torch::Tensor k,
   {
     const long rows = q.numel() / D;
     const long work = rows * (D / 16);
-    hipLaunchKernelGGL(rope_kernel, dim3(grid_for(work, 256)), dim3(256), 0,
-                       stream, (const short*)q.data_ptr(),
-                       (short*)qo.data_ptr(), cosc.data_ptr<float>(),
-                       sinc.data_ptr<float>(), rows, (int)q.size(2), D, S);
+    FP16_SWITCH(f16,
+        hipLaunchKernelGGL((rope_kernel<kFP16>), dim3(grid_for(work, 256)),
+                           dim3(256), 0, stream,
+                           (const short*)q.data_ptr(),
+                           (short*)qo.data_ptr(), cosc.data_ptr<float>(),
+                           sinc.data_ptr<float>(), rows, (int)q.size(2), D,
+                           S));
   }
   {
     const long rows = k.numel() / D;
     const long work = rows * (D / 16);
-    hipLaunchKernelGGL(rope_kernel, dim3(grid_for(work, 256)), dim3(256), 0,
-                       stream, (const short*)k.data_ptr(),
-                       (short*)ko.data_ptr(), cosc.data_ptr<float>(),
-                       sinc.data_ptr<float>(), rows, (int)k.size(2), D, S);
+    FP16_SWITCH(f16,
+        hipLaunchKernelGGL((rope_kernel<kFP16>), dim3(grid_for(work, 256)),
+                           dim3(256), 0, stream,
+                           (const short*)k.data_ptr(),
+                           (short*)ko.data_ptr(), cosc.data_ptr<float>(),
+                           sinc.data_ptr<float>(), rows, (int)k.size(2), D,
+                           S));
   }
   HIP_CHECK_LAST();
   return {qo, ko};
 }
 
 torch::Tensor swiglu_forward(torch::Tensor g, torch::Tensor u) {
-  TORCH_CHECK(g.is_cuda() && g.scalar_type() == torch::kBFloat16,
-              "swiglu: bf16 only on GPU");
+  TORCH_CHECK(g.is_cuda() && (g.scalar_type() == torch::kBFloat16 ||
+                              g.scalar_type() == torch::kHalf),
+              "swiglu: bf16/fp16 only on GPU");
+  const bool f16 = g.scalar_type() == torch::kHalf;
   TORCH_CHECK(g.numel() % 8 == 0);
   auto y = torch::empty_like(g);
   auto stream = at::hip::getCurrentHIPStream();
   const long n8 = g.numel() / 8;
   if (g.numel() % 16 == 0) {
     const long n16 = g.numel() / 16;
-    hipLaunchKernelGGL(swiglu_fwd_kernel_v2, dim3(grid_for(n16, 256)),
-                       dim3(256), 0, stream, (const short*)g.data_ptr(),
-                       (const short*)u.data_ptr(), (short*)y.data_ptr(),
-                       n16);
+    FP16_SWITCH(f16,
+        hipLaunchKernelGGL((swiglu_fwd_kernel_v2<kFP16>),
+                           dim3(grid_for(n16, 256)), dim3(256), 0, stream,
+                           (const short*)g.data_ptr(),
+                           (const short*)u.data_ptr(),
+                           (short*)y.data_ptr(), n16));
   } else {
-    hipLaunchKernelGGL(swiglu_fwd_kernel, dim3(grid_for(n8, 256)),
-                       dim3(256), 0, stream, (const short*)g.data_ptr(),
-                       (const short*)u.data_ptr(), (short*)y.data_ptr(),
-                       n8);
+    FP16_SWITCH(f16,
+        hipLaunchKernelGGL((swiglu_fwd_kernel<kFP16>),
+                           dim3(grid_for(n8, 256)), dim3(256), 0, stream,
+                           (const short*)g.data_ptr(),
+                           (const short*)u.data_ptr(),
+                           (short*)y.data_ptr(), n8));
   }
   HIP_CHECK_LAST();
   return y;
@@ -643,24 +678,32 @@ torch::Tensor swiglu_forward(torch::Tensor g, torch::Tensor u) {
 
 std::vector<torch::Tensor> swiglu_backward(torch::Tensor dy, torch::Tensor g,
                                            torch::Tensor u) {
-  TORCH_CHECK(g.is_cuda() && g.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(g.is_cuda() && (g.scalar_type() == torch::kBFloat16 ||
+                              g.scalar_type() == torch::kHalf));
+  const bool f16 = g.scalar_type() == torch::kHalf;
   auto dg = torch::empty_like(g);
   auto du = torch::empty_like(u);
   auto stream = at::hip::getCurrentHIPStream();
   const long n8 = g.numel() / 8;
   if (dy.numel() % 16 == 0) {
     const long n16 = dy.numel() / 16;
-    hipLaunchKernelGGL(swiglu_bwd_kernel_v2, dim3(grid_for(n16, 256)),
-                       dim3(256), 0, stream, (const short*)dy.data_ptr(),
-                       (const short*)g.data_ptr(),
-                       (const short*)u.data_ptr(), (short*)dg.data_ptr(),
-                       (short*)du.data_ptr(), n16);
+    FP16_SWITCH(f16,
+        hipLaunchKernelGGL((swiglu_bwd_kernel_v2<kFP16>),
+                           dim3(grid_for(n16, 256)), dim3(256), 0, stream,
+                           (const short*)dy.data_ptr(),
+                           (const short*)g.data_ptr(),
+                           (const short*)u.data_ptr(),
+                           (short*)dg.data_ptr(), (short*)du.data_ptr(),
+                           n16));
   } else {
-    hipLaunchKernelGGL(swiglu_bwd_kernel, dim3(grid_for(n8, 256)),
-                       dim3(256), 0, stream, (const short*)dy.data_ptr(),
-                       (const short*)g.data_ptr(),
-                       (const short*)u.data_ptr(), (short*)dg.data_ptr(),
-                       (short*)du.data_ptr(), n8);
+    FP16_SWITCH(f16,
+        hipLaunchKernelGGL((swiglu_bwd_kernel<kFP16>),
+                           dim3(grid_for(n8, 256)), dim3(256), 0, stream,
+                           (const short*)dy.data_ptr(),
+                           (const short*)g.data_ptr(),
+                           (const short*)u.data_ptr(),
+                           (short*)dg.data_ptr(), (short*)du.data_ptr(),
+                           n8));
   }
   HIP_CHECK_LAST();
   return {dg, du};
@@ -671,8 +714,10 @@ std::vector<torch::Tensor> add_rmsnorm_forward(torch::Tensor x,
                                                torch::Tensor resid,
                                                torch::Tensor w, double eps) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous());
-  TORCH_CHECK(x.scalar_type() == torch::kBFloat16,
-              "add_rmsnorm: bf16 only on GPU");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 ||
+                  x.scalar_type() == torch::kHalf,
+              "add_rmsnorm: bf16/fp16 only on GPU");
+  const bool f16 = x.scalar_type() == torch::kHalf;
   const int H = x.size(-1);
   TORCH_CHECK(H % 8 == 0);
   const long rows = x.numel() / H;
@@ -688,23 +733,24 @@ std::vector<torch::Tensor> add_rmsnorm_forward(torch::Tensor x,
   short* rop = (short*)resid_out.data_ptr();
   float* irp = inv_rms.data_ptr<float>();
   dim3 grid(std::min<long>(rows, 2048)), block(256);
-  if (H == 4096) {
-    hipLaunchKernelGGL(add_rmsnorm_fwd_fast<2>, grid, block, 0, stream,
-                       xin, rin, wp, yp, rop, irp, (int)rows, (float)eps,
-                       has_resid);
-  } else if (H == 8192) {
-    hipLaunchKernelGGL(add_rmsnorm_fwd_fast<4>, grid, block, 0, stream,
-                       xin, rin, wp, yp, rop, irp, (int)rows, (float)eps,
-                       has_resid);
-  } else if (H == 2048) {
-    hipLaunchKernelGGL(add_rmsnorm_fwd_fast<1>, grid, block, 0, stream,
-                       xin, rin, wp, yp, rop, irp, (int)rows, (float)eps,
-                       has_resid);
-  } else {
-    hipLaunchKernelGGL(add_rmsnorm_fwd_kernel, grid, block, 0, stream,
-                       xin, rin, wp, yp, rop, irp, (int)rows, H,
-                       (float)eps, has_resid);
-  }
+  FP16_SWITCH(f16,
+      if (H == 4096) {
+        hipLaunchKernelGGL((add_rmsnorm_fwd_fast<2, kFP16>), grid, block, 0,
+                           stream, xin, rin, wp, yp, rop, irp, (int)rows,
+                           (float)eps, has_resid);
+      } else if (H == 8192) {
+        hipLaunchKernelGGL((add_rmsnorm_fwd_fast<4, kFP16>), grid, block, 0,
+                           stream, xin, rin, wp, yp, rop, irp, (int)rows,
+                           (float)eps, has_resid);
+      } else if (H == 2048) {
+        hipLaunchKernelGGL((add_rmsnorm_fwd_fast<1, kFP16>), grid, block, 0,
+                           stream, xin, rin, wp, yp, rop, irp, (int)rows,
+                           (float)eps, has_resid);
+      } else {
+        hipLaunchKernelGGL((add_rmsnorm_fwd_kernel<kFP16>), grid, block, 0,
+                           stream, xin, rin, wp, yp, rop, irp, (int)rows, H,
+                           (float)eps, has_resid);
+      });
   HIP_CHECK_LAST();
   return {y, resid_out, inv_rms};
 }
@@ -728,28 +774,30 @@ std::vector<torch::Tensor> add_rmsnorm_backward(torch::Tensor dy,
   short* dxp = (short*)dx.data_ptr();
   float* dwp = dw32.data_ptr<float>();
   dim3 block(256);
-  if (H == 4096) {
-    hipLaunchKernelGGL(add_rmsnorm_bwd_fast<2>,
-                       dim3(std::min<long>(rows, 512)), block, 0, stream,
-                       dyp, drp, xp, wp, irp, dxp, dwp, (int)rows,
-                       has_dresid);
-  } else if (H == 8192) {
-    hipLaunchKernelGGL(add_rmsnorm_bwd_fast<4>,
-                       dim3(std::min<long>(rows, 512)), block, 0, stream,
-                       dyp, drp, xp, wp, irp, dxp, dwp, (int)rows,
-                       has_dresid);
-  } else if (H == 2048) {
-    hipLaunchKernelGGL(add_rmsnorm_bwd_fast<1>,
-                       dim3(std::min<long>(rows, 512)), block, 0, stream,
-                       dyp, drp, xp, wp, irp, dxp, dwp, (int)rows,
-                       has_dresid);
-  } else {
-    const int lds = (H + 16) * sizeof(float);
-    hipLaunchKernelGGL(add_rmsnorm_bwd_kernel,
-                       dim3(std::min<long>(rows, 512)), block, lds, stream,
-                       dyp, drp, xp, wp, irp, dxp, dwp, (int)rows, H,
-                       has_dresid);
-  }
+  const bool f16 = r_saved.scalar_type() == torch::kHalf;
+  FP16_SWITCH(f16,
+      if (H == 4096) {
+        hipLaunchKernelGGL((add_rmsnorm_bwd_fast<2, kFP16>),
+                           dim3(std::min<long>(rows, 512)), block, 0,
+                           stream, dyp, drp, xp, wp, irp, dxp, dwp,
+                           (int)rows, has_dresid);
+      } else if (H == 8192) {
+        hipLaunchKernelGGL((add_rmsnorm_bwd_fast<4, kFP16>),
+                           dim3(std::min<long>(rows, 512)), block, 0,
+                           stream, dyp, drp, xp, wp, irp, dxp, dwp,
+                           (int)rows, has_dresid);
+      } else if (H == 2048) {
+        hipLaunchKernelGGL((add_rmsnorm_bwd_fast<1, kFP16>),
+                           dim3(std::min<long>(rows, 512)), block, 0,
+                           stream, dyp, drp, xp, wp, irp, dxp, dwp,
+                           (int)rows, has_dresid);
+      } else {
+        const int lds = (H + 16) * sizeof(float);
+        hipLaunchKernelGGL((add_rmsnorm_bwd_kernel<kFP16>),
+                           dim3(std::min<long>(rows, 512)), block, lds,
+                           stream, dyp, drp, xp, wp, irp, dxp, dwp,
+                           (int)rows, H, has_dresid);
+      });
   HIP_CHECK_LAST();
   return {dx, dw32.to(w.scalar_type())};
 }

@@ -23,7 +23,7 @@ class _CrossEntropy(torch.autograd.Function):
     def forward(ctx, logits, target, ignore_index):
         # logits [N, V], target [N]
         ext = dispatch(logits)
-        if logits.dtype != torch.bfloat16:
+        if logits.dtype not in (torch.bfloat16, torch.float16):
             ext = None  # kernel is bf16-only; fp32 uses the composite path
         logits = logits.contiguous()
         if ext is not None:
@@ -46,7 +46,7 @@ class _CrossEntropy(torch.autograd.Function):
     def backward(ctx, dloss):
         logits, target, lse, nvalid = ctx.saved_tensors
         ext = dispatch(logits)
-        if logits.dtype != torch.bfloat16:
+        if logits.dtype not in (torch.bfloat16, torch.float16):
             ext = None
         scale = dloss.float() / nvalid.clamp_min(1).float()
         if ext is not None:
@@ -128,7 +128,7 @@ class _LinearCrossEntropy(torch.autograd.Function):
     def forward(ctx, x, weight, target, ignore_index):
         N = x.shape[0]
         ext = dispatch(x)
-        if x.dtype != torch.bfloat16:
+        if x.dtype not in (torch.bfloat16, torch.float16):
             ext = None
         total = x.new_zeros((), dtype=torch.float32)
         nvalid_t = torch.zeros((), dtype=torch.long, device=x.device)
@@ -150,7 +150,7 @@ class _LinearCrossEntropy(torch.autograd.Function):
         x, weight, target, lse_all, nvalid_t = ctx.saved_tensors
         ignore_index = ctx.ignore_index
         ext = dispatch(x)
-        if x.dtype != torch.bfloat16:
+        if x.dtype not in (torch.bfloat16, torch.float16):
             ext = None
         N = x.shape[0]
         scale = (dloss.float() / nvalid_t.clamp_min(1).float())

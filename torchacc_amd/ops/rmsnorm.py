@@ -21,7 +21,7 @@ class _RMSNorm(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, eps):
         ext = dispatch(x)
-        if x.dtype not in (torch.bfloat16, torch.float32):
+        if x.dtype not in (torch.bfloat16, torch.float16, torch.float32):
             ext = None
         x = x.contiguous()
         if ext is not None:
@@ -37,7 +37,7 @@ class _RMSNorm(torch.autograd.Function):
     def backward(ctx, dy):
         x, weight, inv_rms = ctx.saved_tensors
         ext = dispatch(x)
-        if x.dtype != torch.bfloat16:
+        if x.dtype not in (torch.bfloat16, torch.float16):
             ext = None  # kernel backward is bf16-only
         dy = dy.contiguous()
         if ext is not None:
@@ -83,7 +83,7 @@ class _AddRMSNorm(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, residual, weight, eps):
         ext = dispatch(x)
-        if x.dtype != torch.bfloat16:
+        if x.dtype not in (torch.bfloat16, torch.float16):
             ext = None  # composite path for fp32-on-GPU
         x = x.contiguous()
         has_resid = residual is not None
@@ -105,7 +105,7 @@ class _AddRMSNorm(torch.autograd.Function):
     def backward(ctx, dy, dresid):
         r, weight, inv_rms = ctx.saved_tensors
         ext = dispatch(r)
-        if r.dtype != torch.bfloat16:
+        if r.dtype not in (torch.bfloat16, torch.float16):
             ext = None
         dy = dy.contiguous()
         if ext is not None:

@@ -44,7 +44,7 @@ class _RoPE(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, cos, sin):
         ext = dispatch(q)
-        if q.dtype != torch.bfloat16:
+        if q.dtype not in (torch.bfloat16, torch.float16):
             ext = None
         q = q.contiguous()
         k = k.contiguous()
@@ -63,7 +63,7 @@ class _RoPE(torch.autograd.Function):
         cos = cos.float()
         sin = sin.float()
         ext = dispatch(dq)
-        if dq.dtype != torch.bfloat16:
+        if dq.dtype not in (torch.bfloat16, torch.float16):
             ext = None
         dq = dq.contiguous()
         dk = dk.contiguous()

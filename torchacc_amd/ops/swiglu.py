@@ -13,7 +13,7 @@ class _SwiGLU(torch.autograd.Function):
     @staticmethod
     def forward(ctx, gate, up):
         ext = dispatch(gate)
-        if gate.dtype != torch.bfloat16:
+        if gate.dtype not in (torch.bfloat16, torch.float16):
             ext = None
         gate = gate.contiguous()
         up = up.contiguous()
@@ -27,7 +27,7 @@ class _SwiGLU(torch.autograd.Function):
     def backward(ctx, dy):
         gate, up = ctx.saved_tensors
         ext = dispatch(gate)
-        if gate.dtype != torch.bfloat16:
+        if gate.dtype not in (torch.bfloat16, torch.float16):
             ext = None
         dy = dy.contiguous()
         if ext is not None:
