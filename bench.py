@@ -210,7 +210,7 @@ def main():
                                     cfg.memory.gc_selective_attn
                                     else cfg.memory.gc),
                 "gc_cnt": cfg.memory.gc_cnt,
-                "loss": float(last) if last is not None else None,
+                "loss": float(last.detach()) if last is not None else None,
                 "peak_mem_gb": (round(
                     torch.cuda.max_memory_allocated() / 2**30, 2)
                     if on_gpu else None),
