@@ -209,6 +209,15 @@ class LlamaForCausalLM(nn.Module):
         elif isinstance(m, nn.Embedding):
             nn.init.normal_(m.weight, std=0.02)
 
+    @torch.no_grad()
+    def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
+                 temperature: float = 0.0, top_k: int = 0,
+                 eos_token_id=None) -> torch.Tensor:
+        """KV-cache autoregressive decode (models/generation.py)."""
+        from .generation import generate as _gen
+        return _gen(self, input_ids, max_new_tokens, temperature, top_k,
+                    eos_token_id)
+
     def forward(self, input_ids: torch.Tensor,
                 labels: Optional[torch.Tensor] = None,
                 attention_mask=None):

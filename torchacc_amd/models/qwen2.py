@@ -124,6 +124,15 @@ class Qwen2ForCausalLM(nn.Module):
         if isinstance(m, nn.Linear) and m.bias is not None:
             nn.init.zeros_(m.bias)
 
+    @torch.no_grad()
+    def generate(self, input_ids, max_new_tokens: int = 32,
+                 temperature: float = 0.0, top_k: int = 0,
+                 eos_token_id=None):
+        """KV-cache autoregressive decode (models/generation.py)."""
+        from .generation import generate as _gen
+        return _gen(self, input_ids, max_new_tokens, temperature, top_k,
+                    eos_token_id)
+
     def forward(self, input_ids: torch.Tensor,
                 labels: Optional[torch.Tensor] = None,
                 attention_mask=None):
