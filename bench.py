@@ -42,6 +42,8 @@ def parse_args():
                         "of retaining attention outputs (selective AC)")
     p.add_argument("--no-gc", action="store_true",
                    help="disable gradient checkpointing")
+    p.add_argument("--fp16", action="store_true",
+                   help="train in fp16 (native f16 kernels + GradScaler)")
     p.add_argument("--gc-cnt", type=int, default=None,
                    help="checkpoint only the first N layers (288 GB HBM3E "
                         "rarely needs all of them)")
@@ -83,7 +85,8 @@ def main():
     import torchacc_amd as ta
 
     cfg = ta.Config()
-    cfg.compute.bf16 = on_gpu
+    cfg.compute.bf16 = on_gpu and not args.fp16
+    cfg.compute.fp16 = on_gpu and args.fp16
     parallelism = f"fsdp{world}"
     if args.mode == "fsdp":
         cfg.dist.fsdp.size = world
@@ -127,6 +130,7 @@ def main():
         model, mcfg = build_model(args, cfg)
     model = ta.accelerate(model, config=cfg)
     opt = ta.ops.AdamW(model.parameters(), lr=1e-4, weight_decay=0.0)
+    scaler = ta.amp.GradScaler() if cfg.compute.fp16 else None
 
     device = ta.lazy_device()
     seq = args.seq_len
@@ -149,8 +153,13 @@ def main():
         it[0] += 1
         # labels are shifted inside the model (predict t+1 from t)
         loss = model(ids, labels=ids)
-        loss.backward()
-        opt.step()
+        if scaler is not None:
+            scaler.scale(loss).backward()
+            scaler.step(opt)
+            scaler.update()
+        else:
+            loss.backward()
+            opt.step()
         opt.zero_grad(set_to_none=True)
         return loss
 
@@ -198,7 +207,8 @@ def main():
             "higher_is_better": True,
             "scaling": "strong" if is_cp else "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if on_gpu else "fp32",
+            "dtype": ("fp16" if args.fp16 else "bf16") if on_gpu
+                     else "fp32",
             "data": "synthetic",
             "config": {
                 "model": args.model,
