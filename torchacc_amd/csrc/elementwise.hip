@@ -613,6 +613,8 @@ std::vector<torch::Tensor> rope_forward(torch::Tensor q, torch::Tensor k,
               "rope: bf16/fp16 only on GPU");
   const bool f16 = q.scalar_type() == torch::kHalf;
   TORCH_CHECK(cos.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(cos.is_cuda() && sin.is_cuda(),
+              "rope: cos/sin tables must be on the GPU");
   const int D = q.size(-1);
   TORCH_CHECK(D % 16 == 0, "rope: head_dim must be a multiple of 16");
   const int S = q.size(1);

@@ -200,6 +200,14 @@ class LlamaForCausalLM(nn.Module):
                                     cfg.rope_theta)
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
+
+    def _apply(self, fn, recurse=True):
+        out = super()._apply(fn, recurse)
+        # .to(bfloat16)/.half() sweeps buffers too: the RoPE tables must
+        # stay fp32 (FA2 semantics) — re-float after any cast/move
+        self.rope_cos = self.rope_cos.float()
+        self.rope_sin = self.rope_sin.float()
+        return out
         self.apply(self._init_weights)
 
     @staticmethod

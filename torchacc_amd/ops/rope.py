@@ -17,8 +17,11 @@ from ._backend import dispatch
 
 
 def build_rope_cache(seq_len: int, head_dim: int, base: float = 10000.0,
-                     device="cpu", dtype=torch.float32
+                     device=None, dtype=torch.float32
                      ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """``device=None`` follows the ambient default device (so model
+    construction under ``with torch.device("cuda")`` keeps the tables on
+    the same device as the parameters)."""
     inv_freq = 1.0 / (base ** (
         torch.arange(0, head_dim, 2, device=device, dtype=torch.float32)
         / head_dim))
@@ -48,8 +51,10 @@ class _RoPE(torch.autograd.Function):
             ext = None
         q = q.contiguous()
         k = k.contiguous()
-        cos = cos.float()
-        sin = sin.float()
+        # fp32 tables on q's device: a stale CPU (or dtype-cast) cache must
+        # never reach the kernel as a host pointer / degraded precision
+        cos = cos.to(q.device, torch.float32)
+        sin = sin.to(q.device, torch.float32)
         if ext is not None:
             qo, ko = ext.rope_forward(q, k, cos, sin)
         else:
@@ -60,8 +65,8 @@ class _RoPE(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dq, dk):
         cos, sin = ctx.saved_tensors
-        cos = cos.float()
-        sin = sin.float()
+        cos = cos.to(dq.device, torch.float32)
+        sin = sin.to(dq.device, torch.float32)
         ext = dispatch(dq)
         if dq.dtype not in (torch.bfloat16, torch.float16):
             ext = None
