@@ -116,6 +116,8 @@ class Qwen2ForCausalLM(nn.Module):
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
 
+        self.apply(self._init_weights)
+
     def _apply(self, fn, recurse=True):
         out = super()._apply(fn, recurse)
         # .to(bfloat16)/.half() sweeps buffers too: the RoPE tables must
@@ -123,7 +125,6 @@ class Qwen2ForCausalLM(nn.Module):
         self.rope_cos = self.rope_cos.float()
         self.rope_sin = self.rope_sin.float()
         return out
-        self.apply(self._init_weights)
 
     @staticmethod
     def _init_weights(m):
