@@ -1,0 +1,8 @@
+set -x
+cd /root/repo
+mkdir -p gpurun_out
+timeout 1500 python -m pytest tests/ -m gpu -q 2>&1 | tail -3
+timeout 600 python bench.py --steps 8 --warmup 3 > gpurun_out/c16_default.json 2>/dev/null
+tail -1 gpurun_out/c16_default.json
+timeout 600 python bench.py --steps 6 --warmup 2 --fp16 > gpurun_out/c16_fp16.json 2>/dev/null
+tail -1 gpurun_out/c16_fp16.json
