@@ -115,11 +115,12 @@ def main():
                 and args.model in ("llama-2-7b", "llama-3-8b", "qwen2-7b") \
                 and args.mode == "fsdp" \
                 and args.batch_size * args.seq_len <= 32768:
-            # 288 GB HBM3E rarely needs every layer checkpointed: 8 of 32
-            # measures 18.9k tok/s at 193 GB vs 16.6k at 118 GB for all-32
-            # (profiles/r02); headroom stays ~95 GB and shrinks further as
-            # FSDP shards states at N>1
-            gc_cnt = 8
+            # 288 GB HBM3E rarely needs every layer checkpointed: same-box
+            # sweep (profiles/r02) measured 20.6k tok/s at gc_cnt=4 (206 GB)
+            # vs 20.1k at 8 (193 GB) vs 16.6-17.0k at all-32 (118 GB);
+            # headroom stays >80 GB on every 7B-class config and grows at
+            # N>1 as FSDP shards the optimizer states
+            gc_cnt = 4
         cfg.memory.gc_cnt = gc_cnt
 
     if on_gpu:
