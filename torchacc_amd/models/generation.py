@@ -8,6 +8,11 @@ decode steps run single-query attention against the preallocated KV cache
 applied at absolute positions via table slices, and the logits come from
 one lm_head GEMM over the last position only.
 
+Decode currently materializes a contiguous copy of the live cache slice
+per step (the attention kernels take contiguous [b, s, h, d]); a
+batch-strided kernel path / paged KV layout is the next serving
+optimization.
+
 Usage::
 
     model = LlamaForCausalLM(llama_2_7b()).cuda().to(torch.bfloat16)
