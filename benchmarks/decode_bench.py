@@ -23,6 +23,8 @@ def main():
     p.add_argument("--prompt", type=int, default=128)
     p.add_argument("--new", type=int, default=128)
     p.add_argument("--warmup", type=int, default=8)
+    p.add_argument("--graph", action="store_true",
+                   help="hipGraph-captured decode step (greedy)")
     args = p.parse_args()
 
     from torchacc_amd.models import (LlamaForCausalLM, llama_2_7b,
@@ -42,11 +44,20 @@ def main():
     ids = torch.randint(0, mcfg.vocab_size, (args.batch, args.prompt),
                         device=dev)
 
-    model.generate(ids, max_new_tokens=args.warmup)  # warm
-    if on_gpu:
+    if args.graph:
+        from torchacc_amd.models.generation import GraphDecoder
+        dec = GraphDecoder(model, args.batch,
+                           args.prompt + args.new + args.warmup + 2)
+        dec.decode(ids, args.warmup)  # warm + capture
         torch.cuda.synchronize()
-    t0 = time.perf_counter()
-    out = model.generate(ids, max_new_tokens=args.new)
+        t0 = time.perf_counter()
+        out = dec.decode(ids, args.new)
+    else:
+        model.generate(ids, max_new_tokens=args.warmup)  # warm
+        if on_gpu:
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        out = model.generate(ids, max_new_tokens=args.new)
     if on_gpu:
         torch.cuda.synchronize()
     t1 = time.perf_counter()
@@ -58,6 +69,7 @@ def main():
         "model": args.model, "batch": args.batch,
         "prompt": args.prompt, "new": args.new,
         "dtype": "bf16" if on_gpu else "fp32",
+        "graph": args.graph,
     }))
 
 
