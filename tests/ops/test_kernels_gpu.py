@@ -990,3 +990,18 @@ def test_graph_decoder_matches_eager():
     got2 = dec.decode(ids, 12)
     torch.cuda.synchronize()
     assert torch.equal(got2, got)
+
+
+@pytest.mark.parametrize("m", [1, 3, 8])
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float16])
+def test_lt_gemv_parity(m, dtype):
+    """Skinny decode GEMV vs fp32 matmul."""
+    from torchacc_amd.ops._backend import require_extension
+    ext = require_extension()
+    torch.manual_seed(0)
+    x = torch.randn(m, 4096, device="cuda", dtype=dtype)
+    w = torch.randn(512, 4096, device="cuda", dtype=dtype)
+    y = ext.lt_gemv(x, w)
+    ref = x.float() @ w.float().t()
+    err = (y.float() - ref).abs().max() / ref.abs().max()
+    assert float(err) < 2e-2, float(err)

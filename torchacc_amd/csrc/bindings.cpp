@@ -74,6 +74,7 @@ torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, bool ta, bool tb,
                       int64_t algo_index,
                       c10::optional<torch::Tensor> out_opt);
 std::string lt_gemm_algo_name(int64_t algo_index);
+torch::Tensor lt_gemv(torch::Tensor x, torch::Tensor w);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "torchacc_amd CDNA4 (gfx950) kernels";
@@ -97,4 +98,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("algo_index") = -1,
         pybind11::arg("out") = pybind11::none());
   m.def("lt_gemm_algo_name", &lt_gemm_algo_name);
+  m.def("lt_gemv", &lt_gemv);
 }

@@ -80,13 +80,22 @@ def tuned_linear(x: torch.Tensor, weight: torch.Tensor,
     """F.linear drop-in; hipBLASLt with pinned algos on GPU bf16."""
     ext = None if os.environ.get("TA_DISABLE_TUNED_GEMM") == "1" else _load()
     if (ext is None or not x.is_cuda or bias is not None
-            or x.dtype != torch.bfloat16 or weight.dtype != torch.bfloat16):
+            or x.dtype not in (torch.bfloat16, torch.float16)
+            or weight.dtype != x.dtype):
         return F.linear(x, weight, bias)
     shape = x.shape
     x2d = x.reshape(-1, shape[-1])
     if not x2d.is_contiguous():
         x2d = x2d.contiguous()
-    out = _LtLinear.apply(x2d, weight, ext)
+    if x2d.shape[0] <= 8 and not torch.is_grad_enabled() \
+            and x2d.shape[-1] % 8 == 0:
+        # decode-shaped: the wave-per-row GEMV streams W at HBM rate where
+        # hipBLASLt's m<=8 kernels reach ~2.2 TB/s (bf16 AND fp16)
+        out = ext.lt_gemv(x2d, weight)
+    elif x.dtype == torch.bfloat16:
+        out = _LtLinear.apply(x2d, weight, ext)
+    else:
+        return F.linear(x, weight, bias)
     return out.view(*shape[:-1], weight.shape[0])
 
 
