@@ -87,10 +87,12 @@ def tuned_linear(x: torch.Tensor, weight: torch.Tensor,
     x2d = x.reshape(-1, shape[-1])
     if not x2d.is_contiguous():
         x2d = x2d.contiguous()
-    if x2d.shape[0] <= 8 and not torch.is_grad_enabled() \
+    if x2d.shape[0] <= 4 and not torch.is_grad_enabled() \
             and x2d.shape[-1] % 8 == 0:
-        # decode-shaped: the wave-per-row GEMV streams W at HBM rate where
-        # hipBLASLt's m<=8 kernels reach ~2.2 TB/s (bf16 AND fp16)
+        # decode-shaped: the wave-per-row GEMV streams W once and beats
+        # hipBLASLt's skinny kernels at m<=4 (b1 decode 217->225 tok/s,
+        # +38% under hipGraph); at m=8 the per-wave x reloads lose to
+        # hipBLASLt, so larger micro-batches take the lt path
         out = ext.lt_gemv(x2d, weight)
     elif x.dtype == torch.bfloat16:
         out = _LtLinear.apply(x2d, weight, ext)
