@@ -84,3 +84,8 @@ def _bench_fsdp_tp_worker(rank, world):
 def test_bench_fsdp_tp_cpu_rehearsal():
     """The 70B FSDPxTP launch path (BASELINE config 4) at tp2 x fsdp2."""
     run_multiprocess(_bench_fsdp_tp_worker, world_size=4, timeout=600)
+
+
+def test_bench_ring2_cpu_rehearsal():
+    run_multiprocess(_bench_cp_worker, world_size=2, args=("ring",),
+                     timeout=600)
