@@ -111,16 +111,15 @@ def main():
         cfg.memory.gc_cls = {"LlamaDecoderLayer", "Qwen2DecoderLayer"}
         cfg.memory.gc_selective_attn = not args.no_gc_selective
         gc_cnt = args.gc_cnt
-        if gc_cnt is None \
-                and args.model in ("llama-2-7b", "llama-3-8b", "qwen2-7b") \
-                and args.mode == "fsdp" \
+        if gc_cnt is None and args.mode == "fsdp" \
                 and args.batch_size * args.seq_len <= 32768:
             # 288 GB HBM3E rarely needs every layer checkpointed: same-box
-            # sweep (profiles/r02) measured 20.6k tok/s at gc_cnt=4 (206 GB)
-            # vs 20.1k at 8 (193 GB) vs 16.6-17.0k at all-32 (118 GB);
-            # headroom stays >80 GB on every 7B-class config and grows at
-            # N>1 as FSDP shards the optimizer states
-            gc_cnt = 4
+            # sweeps (profiles/r02) measured llama-2-7b at 20.6k tok/s with
+            # gc_cnt=4 (206 GB) vs 16.6-17.0k at all-32 (118 GB). The
+            # big-vocab models regress beyond ~225 GB (allocator
+            # pressure), so they keep 8 checkpointed layers.
+            gc_cnt = {"llama-2-7b": 4, "llama-3-8b": 8,
+                      "qwen2-7b": 8}.get(args.model)
         cfg.memory.gc_cnt = gc_cnt
 
     if on_gpu:
