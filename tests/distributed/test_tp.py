@@ -243,3 +243,34 @@ def test_hf_llama_tp2():
     for _ in range(2):
         rank, ok, got, want = q.get()
         assert ok, f"rank {rank}: HF TP loss {got} != {want}"
+
+
+def _tp_generate_worker(rank, world, q):
+    """KV-cache generate() through a TP-sharded model (distributed
+    serving): tokens must match the unsharded model."""
+    import torchacc_amd as ta
+    from torchacc_amd.models import LlamaForCausalLM, llama_tiny
+    torch.manual_seed(0)
+    ref = LlamaForCausalLM(llama_tiny()).eval()
+    cfg = ta.Config()
+    cfg.dist.tp.size = world
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(llama_tiny())
+    model = ta.accelerate(model, config=cfg)
+    inner = model._get_underlay_model() if hasattr(
+        model, "_get_underlay_model") else model
+    torch.manual_seed(7)
+    ids = torch.randint(0, 1024, (2, 12))
+    got = inner.generate(ids, max_new_tokens=8)
+    want = ref.generate(ids, max_new_tokens=8)
+    q.put((rank, got.tolist(), want.tolist()))
+
+
+def test_tp2_generate_matches_single():
+    import multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    run_multiprocess(_tp_generate_worker, world_size=2, args=(q,))
+    for _ in range(2):
+        rank, got, want = q.get()
+        assert got == want, f"rank {rank} generated tokens diverge"
