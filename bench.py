@@ -113,12 +113,13 @@ def main():
         gc_cnt = args.gc_cnt
         if gc_cnt is None and args.mode == "fsdp" \
                 and args.batch_size * args.seq_len <= 32768:
-            # 288 GB HBM3E rarely needs every layer checkpointed: same-box
-            # sweeps (profiles/r02) measured llama-2-7b at 20.6k tok/s with
-            # gc_cnt=4 (206 GB) vs 16.6-17.0k at all-32 (118 GB). The
-            # big-vocab models regress beyond ~225 GB (allocator
-            # pressure), so they keep 8 checkpointed layers.
-            gc_cnt = {"llama-2-7b": 4, "llama-3-8b": 8,
+            # 288 GB HBM3E rarely needs every layer checkpointed; depth
+            # picked per model from same-box sweeps (profiles/r02):
+            # llama-2-7b 20.6k at gc4 (206 GB) vs 16.6-17.0k at all-32;
+            # llama-3-8b gc4 18.4k > gc8 18.0k (236 GB); qwen2-7b is the
+            # exception — gc4 is 10% SLOWER than gc8 (17.1k vs 19.0k at
+            # 229 GB), so it keeps 8 checkpointed layers.
+            gc_cnt = {"llama-2-7b": 4, "llama-3-8b": 4,
                       "qwen2-7b": 8}.get(args.model)
         cfg.memory.gc_cnt = gc_cnt
 
